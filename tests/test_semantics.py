@@ -1,0 +1,111 @@
+"""Semantic parity: eager Glom vs the independent loop-level oracle (CPU)."""
+
+import pytest
+import torch
+
+from glom_pytorch_amd import Glom
+from conftest import SMALL, SMALL_BATCH, SMALL_ITERS
+from oracle import oracle_forward
+
+
+def _img(b=SMALL_BATCH, size=SMALL["image_size"]):
+    return torch.randn(b, 3, size, size)
+
+
+def _assert_close(a, b, tol=2e-5):
+    assert a.shape == b.shape
+    assert torch.allclose(a, b, rtol=tol, atol=tol), \
+        f"max abs err {(a - b).abs().max().item():.3e}"
+
+
+def test_default_path():
+    m = Glom(**SMALL)
+    img = _img()
+    _assert_close(m(img, iters=SMALL_ITERS),
+                  oracle_forward(m, img, iters=SMALL_ITERS))
+
+
+def test_consensus_self_true():
+    m = Glom(**SMALL, consensus_self=True)
+    img = _img()
+    _assert_close(m(img, iters=SMALL_ITERS),
+                  oracle_forward(m, img, iters=SMALL_ITERS))
+
+
+def test_local_consensus_radius():
+    m = Glom(**SMALL, local_consensus_radius=2)
+    img = _img()
+    _assert_close(m(img, iters=SMALL_ITERS),
+                  oracle_forward(m, img, iters=SMALL_ITERS))
+
+
+def test_return_all_trajectory():
+    m = Glom(**SMALL)
+    img = _img()
+    out = m(img, iters=SMALL_ITERS, return_all=True)
+    ref = oracle_forward(m, img, iters=SMALL_ITERS, return_all=True)
+    assert out.shape == (SMALL_ITERS + 1, SMALL_BATCH, m.num_patches,
+                         SMALL["levels"], SMALL["dim"])
+    _assert_close(out, ref)
+    # trajectory includes the INITIAL state at t=0
+    init = m.init_levels.view(1, 1, SMALL["levels"], SMALL["dim"]).expand_as(out[0])
+    assert torch.equal(out[0], init)
+
+
+def test_stateful_continuation():
+    m = Glom(**SMALL)
+    img1, img2 = _img(), _img()
+    lv1 = m(img1, iters=3)
+    out = m(img2, iters=2, levels=lv1)
+    ref = oracle_forward(m, img2, iters=2, levels=oracle_forward(m, img1, iters=3))
+    _assert_close(out, ref, tol=5e-5)
+    # continuation on the SAME image equals one longer run
+    lv_a = m(img1, iters=2, levels=m(img1, iters=3))
+    lv_b = m(img1, iters=5)
+    _assert_close(lv_a, lv_b, tol=1e-6)
+
+
+def test_odd_iters_and_single_iter():
+    m = Glom(**SMALL)
+    img = _img()
+    for it in (1, 5):
+        _assert_close(m(img, iters=it), oracle_forward(m, img, iters=it))
+
+
+def test_levels_2_minimum():
+    m = Glom(dim=32, levels=2, image_size=16, patch_size=8)
+    img = torch.randn(2, 3, 16, 16)
+    _assert_close(m(img, iters=2), oracle_forward(m, img, iters=2))
+
+
+def test_levels_below_2_rejected():
+    with pytest.raises(ValueError):
+        Glom(dim=32, levels=1, image_size=16, patch_size=8)
+
+
+def test_default_iters_is_2x_levels():
+    m = Glom(**SMALL)
+    img = _img()
+    out_default = m(img, return_all=True)
+    assert out_default.shape[0] == 2 * SMALL["levels"] + 1
+
+
+def test_zero_iters_returns_init():
+    m = Glom(**SMALL)
+    img = _img()
+    out = m(img, iters=0)
+    init = m.init_levels.view(1, 1, SMALL["levels"], SMALL["dim"]).expand_as(out)
+    assert torch.equal(out, init)
+
+
+def test_gradients_flow():
+    m = Glom(**SMALL)
+    img = _img()
+    out = m(img, iters=2, return_all=True)
+    loss = out[1, :, :, -1].pow(2).mean()
+    loss.backward()
+    assert m.init_levels.grad is not None
+    assert m.bottom_up.net[1].weight.grad is not None
+    assert m.top_down.net[3].weight.grad is not None
+    assert m.pos_emb.weight.grad is not None
+    assert m.image_to_tokens[1].weight.grad is not None
