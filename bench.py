@@ -1,0 +1,124 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GLOM denoising training step (fwd + bwd + optimizer).
+
+Metric (BASELINE.json): images/sec (whole node), Glom dim=512 levels=6
+image_size=224 patch_size=14 iters=12, bf16, synthetic 224x224 images,
+random-init weights. Weak scaling: fixed per-GPU batch (default 64).
+
+Single GPU:   python bench.py --gpus 1 --steps K --warmup W
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=64, help="per-GPU batch")
+    p.add_argument("--iters", type=int, default=12)
+    p.add_argument("--dim", type=int, default=512)
+    p.add_argument("--levels", type=int, default=6)
+    p.add_argument("--image-size", type=int, default=224)
+    p.add_argument("--patch-size", type=int, default=14)
+    p.add_argument("--impl", choices=["native", "eager", "compile"],
+                   default="native",
+                   help="native = CDNA4 HIP engine; eager/compile = stock "
+                        "PyTorch-ROCm running the same math (comparison)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from glom_pytorch_amd import Glom
+    from glom_pytorch_amd.parallel.trainer import DenoisingTrainer
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group("nccl")
+    torch.cuda.set_device(local_rank)
+    dev = torch.device("cuda", local_rank)
+    torch.manual_seed(1234 + rank)
+
+    model = Glom(dim=args.dim, levels=args.levels, image_size=args.image_size,
+                 patch_size=args.patch_size).to(dev, torch.bfloat16)
+    if args.impl != "native":
+        model.force_eager = True
+    if args.impl == "compile":
+        model = torch.compile(model)
+
+    trainer = DenoisingTrainer(model, distributed=distributed)
+
+    img = torch.randn(args.batch, 3, args.image_size, args.image_size,
+                      device=dev, dtype=torch.bfloat16)
+
+    def sync_all():
+        if distributed:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        trainer.step(img, iters=args.iters)
+
+    sync_all()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = trainer.step(img, iters=args.iters)
+    sync_all()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=dev, dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    if rank == 0:
+        n_gpus = world if distributed else 1
+        global_batch = args.batch * n_gpus
+        images_sec = global_batch * args.steps / elapsed
+        print(json.dumps({
+            "metric": "images/sec",
+            "value": images_sec,
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": f"glom-{args.dim}x{args.levels}",
+                "global_batch": global_batch,
+                "image_size": args.image_size,
+                "patch_size": args.patch_size,
+                "iters": args.iters,
+                "objective": "denoising-mse fwd+bwd+adamw",
+                "impl": args.impl,
+                "parallelism": f"dp{n_gpus}",
+            },
+            "loss_last": loss,
+        }))
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,383 @@
+// Torch bindings: compose the generic MFMA GEMM + row-wise kernels into the
+// five GLOM hot-path ops (grouped FF fwd/bwd, consensus attention fwd/bwd,
+// level mix fwd/bwd). Layout contracts follow SURVEY.md §2.2/§2.3; the
+// grouped-conv weights (G*mult*d, d, 1)/(G*d, mult*d, 1) are consumed
+// directly as G packed GEMM operands.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <cmath>
+#include <cstring>
+#include <vector>
+
+#include "gemm.h"
+#include "aux_kernels.h"
+
+namespace {
+
+#define CHECK_IN(x)                                                        \
+    TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                        \
+    TORCH_CHECK(x.is_contiguous(), #x " must be contiguous");              \
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16, #x " must be bf16")
+
+hipStream_t cur_stream() {
+    return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_launch() {
+    hipError_t e = hipGetLastError();
+    TORCH_CHECK(e == hipSuccess, "HIP launch failed: ", hipGetErrorString(e));
+}
+
+GemmParams base_params(int64_t M, int64_t N, int64_t K, int layout,
+                       int64_t nproblems, int64_t nInner, float alpha) {
+    GemmParams p;
+    std::memset(&p, 0, sizeof(p));
+    p.M = (int)M; p.N = (int)N; p.K = (int)K;
+    p.layout = layout;
+    p.nproblems = (int)nproblems;
+    p.nInner = (int)nInner;
+    p.alpha = alpha;
+    p.epilogue = EPI_NONE;
+    return p;
+}
+
+// ------------------------------------------------------------------ //
+// grouped feed-forward (bottom-up: mode 0, top-down: mode 1)
+// reference glom_pytorch.py:23-36 (GroupedFeedForward) applied at :134/:136
+
+std::vector<torch::Tensor> grouped_ff_fwd(
+        c10::optional<torch::Tensor> tokens_opt, torch::Tensor levels,
+        c10::optional<torch::Tensor> pos_opt, torch::Tensor w1,
+        torch::Tensor b1, torch::Tensor w2, torch::Tensor b2, int64_t mode) {
+    CHECK_IN(levels); CHECK_IN(w1); CHECK_IN(b1); CHECK_IN(w2); CHECK_IN(b2);
+    const int64_t B = levels.size(0), N = levels.size(1),
+                  L = levels.size(2), d = levels.size(3);
+    const int64_t G = (mode == 0) ? L : L - 1;
+    TORCH_CHECK(G <= GEMM_MAX_TABLE, "levels > 16 unsupported by table mode");
+    const int64_t m4 = w1.size(0) / G;
+    const int64_t M = B * N;
+    TORCH_CHECK(d % 8 == 0 && m4 % 8 == 0, "dim and mult*dim must be /8");
+    TORCH_CHECK(w1.size(0) == G * m4 && w1.size(1) == d);
+    TORCH_CHECK(w2.size(0) == G * d && w2.size(1) == m4);
+
+    auto opts = levels.options();
+    auto Hpre = torch::empty({G, M, m4}, opts);
+    auto Y = torch::empty({B, N, G, d}, opts);
+    hipStream_t s = cur_stream();
+
+    // up-projection: Hpre_g = x_g @ W1_g^T + b1_g
+    {
+        GemmParams p = base_params(M, m4, d, LAYOUT_NT, G, G, 1.0f);
+        if (mode == 0) {
+            TORCH_CHECK(tokens_opt.has_value(), "bottom-up needs tokens");
+            auto tokens = tokens_opt.value();
+            CHECK_IN(tokens);
+            p.A.flags = OP_TABLE;
+            p.Atab[0] = tokens.data_ptr();
+            p.Atabld[0] = d;
+            for (int64_t g = 1; g < G; g++) {
+                p.Atab[g] = (const char*)levels.data_ptr() + (g - 1) * d * 2;
+                p.Atabld[g] = L * d;
+            }
+        } else {
+            TORCH_CHECK(pos_opt.has_value(), "top-down needs pos");
+            auto pos = pos_opt.value();
+            CHECK_IN(pos);
+            p.A.base = (const char*)levels.data_ptr() + d * 2;
+            p.A.sin = d; p.A.sout = 0; p.A.ld = L * d;
+            p.A.flags = OP_POS;
+            p.pos = pos.data_ptr();
+            p.pos_ld = d;
+            p.npatch = (int)N;
+        }
+        p.B.base = w1.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
+        p.Cbase = Hpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
+        p.bias_base = b1.data_ptr(); p.bias_sin = m4; p.has_bias = 1;
+        launch_gemm(p, s); check_launch();
+    }
+    // down-projection: Y_g = gelu(Hpre_g) @ W2_g^T + b2_g
+    {
+        GemmParams p = base_params(M, d, m4, LAYOUT_NT, G, G, 1.0f);
+        p.A.base = Hpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
+        p.A.flags = OP_GELU;
+        p.B.base = w2.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
+        p.Cbase = (char*)Y.data_ptr() ; p.Csin = d; p.Cld = G * d;
+        p.bias_base = b2.data_ptr(); p.bias_sin = d; p.has_bias = 1;
+        launch_gemm(p, s); check_launch();
+    }
+    return {Y, Hpre};
+}
+
+std::vector<torch::Tensor> grouped_ff_bwd(
+        torch::Tensor dY, c10::optional<torch::Tensor> tokens_opt,
+        torch::Tensor levels, c10::optional<torch::Tensor> pos_opt,
+        torch::Tensor w1, torch::Tensor w2, torch::Tensor Hpre,
+        int64_t mode) {
+    CHECK_IN(dY); CHECK_IN(levels); CHECK_IN(w1); CHECK_IN(w2); CHECK_IN(Hpre);
+    const int64_t B = levels.size(0), N = levels.size(1),
+                  L = levels.size(2), d = levels.size(3);
+    const int64_t G = (mode == 0) ? L : L - 1;
+    const int64_t m4 = w1.size(0) / G;
+    const int64_t M = B * N;
+    auto opts = levels.options();
+    hipStream_t s = cur_stream();
+
+    auto dHpre = torch::empty({G, M, m4}, opts);
+    auto dLevels = torch::zeros({B, N, L, d}, opts);
+    torch::Tensor dTokens;
+
+    // dHpre = (dY_g @ W2_g) * gelu'(Hpre_g)
+    {
+        GemmParams p = base_params(M, m4, d, LAYOUT_NN, G, G, 1.0f);
+        p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
+        p.B.base = w2.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
+        p.Cbase = dHpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
+        p.epilogue = EPI_GELUGRAD;
+        p.aux_base = Hpre.data_ptr(); p.aux_sin = M * m4; p.aux_ld = m4;
+        launch_gemm(p, s); check_launch();
+    }
+    // dX_g = dHpre_g @ W1_g, scattered into the level slices (or tokens)
+    {
+        GemmParams p = base_params(M, d, m4, LAYOUT_NN, G, G, 1.0f);
+        p.A.base = dHpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
+        p.B.base = w1.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
+        if (mode == 0) {
+            dTokens = torch::empty({B, N, d}, opts);
+            p.Cflags = OP_TABLE;
+            p.Ctab[0] = dTokens.data_ptr();
+            p.Ctabld[0] = d;
+            for (int64_t g = 1; g < G; g++) {
+                p.Ctab[g] = (char*)dLevels.data_ptr() + (g - 1) * d * 2;
+                p.Ctabld[g] = L * d;
+            }
+        } else {
+            p.Cbase = (char*)dLevels.data_ptr() + d * 2;
+            p.Csin = d; p.Cld = L * d;
+        }
+        launch_gemm(p, s); check_launch();
+    }
+    // dW1_g[h, j] = sum_m dHpre_g[m, h] * x_g[m, j]
+    auto dW1 = torch::empty({G * m4, d}, opts);
+    {
+        GemmParams p = base_params(m4, d, M, LAYOUT_TN, G, G, 1.0f);
+        p.A.base = dHpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
+        if (mode == 0) {
+            auto tokens = tokens_opt.value();
+            CHECK_IN(tokens);
+            p.B.flags = OP_TABLE;
+            p.Btab[0] = tokens.data_ptr();
+            p.Btabld[0] = d;
+            for (int64_t g = 1; g < G; g++) {
+                p.Btab[g] = (const char*)levels.data_ptr() + (g - 1) * d * 2;
+                p.Btabld[g] = L * d;
+            }
+        } else {
+            auto pos = pos_opt.value();
+            CHECK_IN(pos);
+            p.B.base = (const char*)levels.data_ptr() + d * 2;
+            p.B.sin = d; p.B.ld = L * d;
+            p.B.flags = OP_POS;
+            p.pos = pos.data_ptr();
+            p.pos_ld = d;
+            p.npatch = (int)N;
+        }
+        p.Cbase = dW1.data_ptr(); p.Csin = m4 * d; p.Cld = d;
+        launch_gemm(p, s); check_launch();
+    }
+    // dW2_g[o, h] = sum_m dY_g[m, o] * gelu(Hpre_g[m, h])
+    auto dW2 = torch::empty({G * d, m4}, opts);
+    {
+        GemmParams p = base_params(d, m4, M, LAYOUT_TN, G, G, 1.0f);
+        p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
+        p.B.base = Hpre.data_ptr(); p.B.sin = M * m4; p.B.ld = m4;
+        p.B.flags = OP_GELU;
+        p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
+        launch_gemm(p, s); check_launch();
+    }
+    auto dB1 = dHpre.sum(1).flatten();                       // (G*m4)
+    auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();       // (G*d)
+    if (mode != 0) dTokens = torch::empty({0}, opts);
+    return {dTokens, dLevels, dW1, dB1, dW2, dB2};
+}
+
+// ------------------------------------------------------------------ //
+// consensus attention (reference glom_pytorch.py:38-73)
+
+std::vector<torch::Tensor> consensus_fwd(
+        torch::Tensor levels, bool attend_self,
+        c10::optional<torch::Tensor> mask_opt) {
+    CHECK_IN(levels);
+    const int64_t B = levels.size(0), N = levels.size(1),
+                  L = levels.size(2), d = levels.size(3);
+    const int64_t P = B * L;
+    auto opts = levels.options();
+    hipStream_t s = cur_stream();
+    const bool* mask = nullptr;
+    if (mask_opt.has_value()) {
+        auto m = mask_opt.value();
+        TORCH_CHECK(m.is_cuda() && m.is_contiguous()
+                    && m.scalar_type() == at::kBool);
+        mask = m.data_ptr<bool>();
+    }
+
+    auto rnorm = torch::empty({B, L, N}, opts.dtype(at::kFloat));
+    launch_rnorm(levels.data_ptr(), rnorm.data_ptr<float>(),
+                 (int)B, (int)N, (int)L, (int)d, s);
+    check_launch();
+
+    auto probs = torch::empty({B, L, N, N}, opts);
+    // scores[i,j] = (q_i . k_j) * rnorm_j * d^-0.5
+    {
+        GemmParams p = base_params(N, N, d, LAYOUT_NT, P, L,
+                                   (float)std::pow((double)d, -0.5));
+        p.A.base = levels.data_ptr(); p.A.sin = d; p.A.sout = N * L * d;
+        p.A.ld = L * d;
+        p.B = p.A;
+        p.Cbase = probs.data_ptr(); p.Csin = N * N; p.Csout = L * N * N;
+        p.Cld = N;
+        p.colscale_base = rnorm.data_ptr<float>();
+        p.cs_sin = N; p.cs_sout = L * N; p.has_colscale = 1;
+        launch_gemm(p, s); check_launch();
+    }
+    // masked softmax in place
+    launch_softmax_fwd(probs.data_ptr(), probs.data_ptr(), mask, (int)P,
+                       (int)N, attend_self ? 0 : 1, s);
+    check_launch();
+    // out[i,:] = sum_j P[i,j] * levels[j,:]
+    auto out = torch::empty({B, N, L, d}, opts);
+    {
+        GemmParams p = base_params(N, d, N, LAYOUT_NN, P, L, 1.0f);
+        p.A.base = probs.data_ptr(); p.A.sin = N * N; p.A.sout = L * N * N;
+        p.A.ld = N;
+        p.B.base = levels.data_ptr(); p.B.sin = d; p.B.sout = N * L * d;
+        p.B.ld = L * d;
+        p.Cbase = out.data_ptr(); p.Csin = d; p.Csout = N * L * d;
+        p.Cld = L * d;
+        launch_gemm(p, s); check_launch();
+    }
+    return {out, probs, rnorm};
+}
+
+torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
+                            torch::Tensor probs, torch::Tensor rnorm,
+                            bool attend_self,
+                            c10::optional<torch::Tensor> mask_opt) {
+    CHECK_IN(dOut); CHECK_IN(levels); CHECK_IN(probs);
+    const int64_t B = levels.size(0), N = levels.size(1),
+                  L = levels.size(2), d = levels.size(3);
+    const int64_t P = B * L;
+    auto opts = levels.options();
+    hipStream_t s = cur_stream();
+    const bool* mask = nullptr;
+    if (mask_opt.has_value()) mask = mask_opt.value().data_ptr<bool>();
+
+    // dP[i,j] = dOut_i . v_j
+    auto dP = torch::empty({B, L, N, N}, opts);
+    {
+        GemmParams p = base_params(N, N, d, LAYOUT_NT, P, L, 1.0f);
+        p.A.base = dOut.data_ptr(); p.A.sin = d; p.A.sout = N * L * d;
+        p.A.ld = L * d;
+        p.B.base = levels.data_ptr(); p.B.sin = d; p.B.sout = N * L * d;
+        p.B.ld = L * d;
+        p.Cbase = dP.data_ptr(); p.Csin = N * N; p.Csout = L * N * N;
+        p.Cld = N;
+        launch_gemm(p, s); check_launch();
+    }
+    auto dS = torch::empty_like(dP);
+    auto dSr = torch::empty_like(dP);
+    launch_softmax_bwd(probs.data_ptr(), dP.data_ptr(),
+                       rnorm.data_ptr<float>(), dS.data_ptr(), dSr.data_ptr(),
+                       mask, (int)P, (int)N, attend_self ? 0 : 1,
+                       (float)std::pow((double)d, -0.5), s);
+    check_launch();
+
+    // dv[j,:] = sum_i P[i,j] dOut[i,:]
+    auto dv = torch::empty({B, N, L, d}, opts);
+    {
+        GemmParams p = base_params(N, d, N, LAYOUT_TN, P, L, 1.0f);
+        p.A.base = probs.data_ptr(); p.A.sin = N * N; p.A.sout = L * N * N;
+        p.A.ld = N;
+        p.B.base = dOut.data_ptr(); p.B.sin = d; p.B.sout = N * L * d;
+        p.B.ld = L * d;
+        p.Cbase = dv.data_ptr(); p.Csin = d; p.Csout = N * L * d;
+        p.Cld = L * d;
+        launch_gemm(p, s); check_launch();
+    }
+    // dq[i,:] = sum_j dSr[i,j] levels[j,:]
+    auto dq = torch::empty({B, N, L, d}, opts);
+    {
+        GemmParams p = base_params(N, d, N, LAYOUT_NN, P, L, 1.0f);
+        p.A.base = dSr.data_ptr(); p.A.sin = N * N; p.A.sout = L * N * N;
+        p.A.ld = N;
+        p.B.base = levels.data_ptr(); p.B.sin = d; p.B.sout = N * L * d;
+        p.B.ld = L * d;
+        p.Cbase = dq.data_ptr(); p.Csin = d; p.Csout = N * L * d;
+        p.Cld = L * d;
+        launch_gemm(p, s); check_launch();
+    }
+    // dkhat[j,:] = sum_i dS[i,j] levels[i,:]   (layout (B,L,N,d))
+    auto dkhat = torch::empty({B, L, N, d}, opts);
+    {
+        GemmParams p = base_params(N, d, N, LAYOUT_TN, P, L, 1.0f);
+        p.A.base = dS.data_ptr(); p.A.sin = N * N; p.A.sout = L * N * N;
+        p.A.ld = N;
+        p.B.base = levels.data_ptr(); p.B.sin = d; p.B.sout = N * L * d;
+        p.B.ld = L * d;
+        p.Cbase = dkhat.data_ptr(); p.Csin = N * d; p.Csout = L * N * d;
+        p.Cld = d;
+        launch_gemm(p, s); check_launch();
+    }
+    auto dLevels = torch::empty({B, N, L, d}, opts);
+    launch_knorm_combine(dkhat.data_ptr(), levels.data_ptr(),
+                         rnorm.data_ptr<float>(), dv.data_ptr(),
+                         dq.data_ptr(), dLevels.data_ptr(),
+                         (int)B, (int)N, (int)L, (int)d, s);
+    check_launch();
+    return dLevels;
+}
+
+// ------------------------------------------------------------------ //
+// level mixing (reference glom_pytorch.py:128-144)
+
+torch::Tensor level_mix_fwd(torch::Tensor levels, torch::Tensor bu,
+                            torch::Tensor td, torch::Tensor cons) {
+    CHECK_IN(levels); CHECK_IN(bu); CHECK_IN(td); CHECK_IN(cons);
+    const int64_t L = levels.size(2), d = levels.size(3);
+    auto out = torch::empty_like(levels);
+    launch_mix_fwd(levels.data_ptr(), bu.data_ptr(), td.data_ptr(),
+                   cons.data_ptr(), out.data_ptr(), levels.numel(), (int)L,
+                   (int)d, cur_stream());
+    check_launch();
+    return out;
+}
+
+std::vector<torch::Tensor> level_mix_bwd(torch::Tensor dout) {
+    CHECK_IN(dout);
+    const int64_t B = dout.size(0), N = dout.size(1), L = dout.size(2),
+                  d = dout.size(3);
+    auto dmix = torch::empty_like(dout);
+    auto dtd = torch::empty({B, N, L - 1, d}, dout.options());
+    launch_mix_bwd(dout.data_ptr(), dmix.data_ptr(), dtd.data_ptr(),
+                   dout.numel(), (int)L, (int)d, cur_stream());
+    check_launch();
+    return {dmix, dtd};
+}
+
+std::string build_info() {
+    return "glom_pytorch_amd HIP extension (gfx950, bf16 MFMA 16x16x32)";
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("grouped_ff_fwd", &grouped_ff_fwd, "grouped FF forward");
+    m.def("grouped_ff_bwd", &grouped_ff_bwd, "grouped FF backward");
+    m.def("consensus_fwd", &consensus_fwd, "consensus attention forward");
+    m.def("consensus_bwd", &consensus_bwd, "consensus attention backward");
+    m.def("level_mix_fwd", &level_mix_fwd, "level mix forward");
+    m.def("level_mix_bwd", &level_mix_bwd, "level mix backward");
+    m.def("build_info", &build_info);
+}

@@ -1,0 +1,45 @@
+// Shared host/device parameter block for the generic GLOM GEMM.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+enum GemmLayout { LAYOUT_NT = 0, LAYOUT_NN = 1, LAYOUT_TN = 2 };
+enum OpFlags { OP_GELU = 1, OP_POS = 2, OP_TABLE = 4 };
+enum GemmEpilogue { EPI_NONE = 0, EPI_GELUGRAD = 1 };
+
+#define GEMM_MAX_TABLE 16
+
+// Per-operand addressing: strided mode resolves problem p's base pointer as
+//   base + (p % nInner) * sin + (p / nInner) * sout      (element strides)
+// or, when OP_TABLE is set, from the explicit pointer table (<=16 problems).
+struct OpArg {
+    const void* base;
+    long sin, sout;
+    long ld;     // leading (row) stride in elements
+    int flags;   // OP_GELU | OP_POS | OP_TABLE
+};
+
+struct GemmParams {
+    int M, N, K;
+    int layout;          // GemmLayout
+    int nproblems, nInner;
+    int npatch;          // pos-emb row period for OP_POS
+    int epilogue;        // GemmEpilogue
+    float alpha;
+
+    OpArg A, B;
+
+    void* Cbase; long Csin, Csout, Cld; int Cflags;
+
+    const void* bias_base; long bias_sin, bias_sout; int has_bias;
+    const void* colscale_base; long cs_sin, cs_sout; int has_colscale;
+    const void* aux_base; long aux_sin, aux_sout, aux_ld;
+
+    const void* pos; long pos_ld;
+
+    const void* Atab[GEMM_MAX_TABLE]; long Atabld[GEMM_MAX_TABLE];
+    const void* Btab[GEMM_MAX_TABLE]; long Btabld[GEMM_MAX_TABLE];
+    void* Ctab[GEMM_MAX_TABLE]; long Ctabld[GEMM_MAX_TABLE];
+};
+
+void launch_gemm(const GemmParams& p, hipStream_t stream);

@@ -1,0 +1,120 @@
+"""Autograd wrappers around the CDNA4 HIP ops + the native Glom forward.
+
+Every hot op of the reference forward is a custom
+torch.autograd.Function with hand-written HIP forward AND backward kernels;
+autograd only stitches the iteration loop together (and re-traverses it for
+losses attached at any (time, level) of the trajectory, as in the
+reference's denoising recipe, README.md:56-90).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from glom_pytorch_amd.ops import _load_extension
+
+
+class GroupedFFFn(torch.autograd.Function):
+    """Grouped per-level MLP (d -> 4d -> GELU -> d), bottom-up or top-down.
+
+    mode 0 (bottom-up): group 0 consumes `tokens`, group g consumes
+    levels[..., g-1, :]  (reference glom_pytorch.py:132-134, without the cat)
+    mode 1 (top-down): group g consumes levels[..., g+1, :] + pos, fused into
+    the GEMM A-operand load (reference glom_pytorch.py:136).
+    """
+
+    @staticmethod
+    def forward(ctx, tokens, levels, pos, w1, b1, w2, b2, mode):
+        ext = _load_extension()
+        Y, Hpre = ext.grouped_ff_fwd(tokens, levels, pos, w1, b1, w2, b2,
+                                     mode)
+        ctx.save_for_backward(tokens, levels, pos, w1, w2, Hpre)
+        ctx.mode = mode
+        return Y
+
+    @staticmethod
+    def backward(ctx, dY):
+        ext = _load_extension()
+        tokens, levels, pos, w1, w2, Hpre = ctx.saved_tensors
+        dTokens, dLevels, dW1, dB1, dW2, dB2 = ext.grouped_ff_bwd(
+            dY.contiguous(), tokens, levels, pos, w1, w2, Hpre, ctx.mode)
+        dPos = None
+        if ctx.mode == 1 and pos is not None and ctx.needs_input_grad[2]:
+            # pos was added to every top-down group input; its grad is the
+            # sum of the level-slice grads over batch and groups.
+            dPos = dLevels[:, :, 1:, :].sum(dim=(0, 2))
+        if ctx.mode == 0:
+            return dTokens, dLevels, None, dW1, dB1, dW2, dB2, None
+        return None, dLevels, dPos, dW1, dB1, dW2, dB2, None
+
+
+class ConsensusFn(torch.autograd.Function):
+    """Consensus attention across patch columns (glom_pytorch.py:38-73)."""
+
+    @staticmethod
+    def forward(ctx, levels, attend_self, mask):
+        ext = _load_extension()
+        out, probs, rnorm = ext.consensus_fwd(levels, attend_self, mask)
+        ctx.save_for_backward(levels, probs, rnorm, mask)
+        ctx.attend_self = attend_self
+        return out
+
+    @staticmethod
+    def backward(ctx, dOut):
+        ext = _load_extension()
+        levels, probs, rnorm, mask = ctx.saved_tensors
+        dLevels = ext.consensus_bwd(dOut.contiguous(), levels, probs, rnorm,
+                                    ctx.attend_self, mask)
+        return dLevels, None, None
+
+
+class LevelMixFn(torch.autograd.Function):
+    """(prev + bu + pad(td) + cons) / [4,..,4,3]  (glom_pytorch.py:141-144)."""
+
+    @staticmethod
+    def forward(ctx, levels, bu, td, cons):
+        ext = _load_extension()
+        return ext.level_mix_fwd(levels, bu, td, cons)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _load_extension()
+        dmix, dtd = ext.level_mix_bwd(dout.contiguous())
+        # prev/bu/cons share the same scaled gradient; consumers only read
+        # their incoming grads, so aliasing one tensor three times is safe.
+        return dmix, dmix, dtd, dmix
+
+
+def glom_step(model, tokens, levels, pos, mask):
+    w = model.bottom_up.net
+    bu = GroupedFFFn.apply(tokens, levels, None, w[1].weight[..., 0],
+                           w[1].bias, w[3].weight[..., 0], w[3].bias, 0)
+    w = model.top_down.net
+    td = GroupedFFFn.apply(None, levels, pos, w[1].weight[..., 0],
+                           w[1].bias, w[3].weight[..., 0], w[3].bias, 1)
+    cons = ConsensusFn.apply(levels, model.attention.attend_self, mask)
+    return LevelMixFn.apply(levels, bu, td, cons)
+
+
+def glom_forward(model, img, iters, levels=None, return_all=False):
+    b = img.shape[0]
+    tokens = model.image_to_tokens(img)          # K1: once per forward
+    n = tokens.shape[1]
+    pos = model.pos_emb.weight
+    mask = (model.attention.non_local_mask
+            if model.attention.local_consensus_radius > 0 else None)
+
+    if levels is None:
+        levels = model.init_levels.view(1, 1, model.levels, model.dim) \
+            .expand(b, n, model.levels, model.dim).contiguous()
+    else:
+        levels = levels.contiguous()
+
+    trajectory = [levels]
+    for _ in range(iters):
+        levels = glom_step(model, tokens, levels, pos, mask)
+        trajectory.append(levels)
+
+    if return_all:
+        return torch.stack(trajectory)
+    return levels

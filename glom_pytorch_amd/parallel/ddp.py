@@ -1,0 +1,92 @@
+"""Bucketed data-parallel gradient all-reduce over RCCL/xGMI.
+
+A deliberately thin DDP built for the MI355X node topology: each GPU has 7
+point-to-point xGMI links (~153 GB/s each), so ring all-reduce is per-link
+bound and the right shape is a handful of medium buckets (default 16 MB)
+launched asynchronously on the communication stream as soon as their
+gradients are produced — overlapping the all-reduce of late-iteration
+gradients with the backward of earlier iterations.
+
+GLOM's 23.5 M params (47 MB bf16) fit in ~3 buckets; reverse registration
+order approximates backward completion order.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+
+class BucketedDDP:
+    def __init__(self, module: torch.nn.Module, bucket_bytes: int = 16 << 20,
+                 process_group=None):
+        self.module = module
+        self.pg = process_group
+        self.world = dist.get_world_size(process_group)
+        # every rank starts from rank 0's weights
+        with torch.no_grad():
+            for p in module.parameters():
+                dist.broadcast(p.data, 0, group=process_group)
+
+        params = [p for p in module.parameters() if p.requires_grad]
+        self.buckets: list[list[torch.nn.Parameter]] = []
+        cur, cur_bytes = [], 0
+        for p in reversed(params):   # ~backward completion order
+            cur.append(p)
+            cur_bytes += p.numel() * p.element_size()
+            if cur_bytes >= bucket_bytes:
+                self.buckets.append(cur)
+                cur, cur_bytes = [], 0
+        if cur:
+            self.buckets.append(cur)
+
+        self._bucket_of = {}
+        for bi, b in enumerate(self.buckets):
+            for p in b:
+                self._bucket_of[p] = bi
+        self._pending = [0] * len(self.buckets)
+        self._works: list = []
+        self._flats: list = []
+        self._reset_counts()
+
+        self._hooks = [
+            p.register_post_accumulate_grad_hook(self._on_grad)
+            for p in params
+        ]
+
+    def _reset_counts(self):
+        for bi, b in enumerate(self.buckets):
+            self._pending[bi] = len(b)
+
+    def _on_grad(self, p):
+        bi = self._bucket_of[p]
+        self._pending[bi] -= 1
+        if self._pending[bi] == 0:
+            self._launch(bi)
+
+    def _launch(self, bi):
+        grads = [p.grad for p in self.buckets[bi]]
+        flat = torch._utils._flatten_dense_tensors(grads)
+        work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=self.pg,
+                               async_op=True)
+        self._works.append(work)
+        self._flats.append((bi, flat))
+
+    def finalize(self):
+        """Wait for all in-flight all-reduces and write averaged grads back.
+
+        Call between loss.backward() and optimizer.step().
+        """
+        for w in self._works:
+            w.wait()
+        inv = 1.0 / self.world
+        for bi, flat in self._flats:
+            flat.mul_(inv)
+            grads = [p.grad for p in self.buckets[bi]]
+            for g, synced in zip(grads,
+                                 torch._utils._unflatten_dense_tensors(
+                                     flat, grads)):
+                g.copy_(synced)
+        self._works.clear()
+        self._flats.clear()
+        self._reset_counts()

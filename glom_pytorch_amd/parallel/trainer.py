@@ -1,0 +1,105 @@
+"""Denoising self-supervised trainer for GLOM (the reference's training
+recipe, README.md:56-90): noise the image, run the model with return_all,
+decode the top level at a mid-iteration timestep back to pixels, MSE against
+the clean image. Adds what the reference leaves to the user: optimizer,
+data-parallel gradient sync, checkpointing, metrics."""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+from einops.layers.torch import Rearrange
+from torch import nn
+
+from glom_pytorch_amd.parallel.ddp import BucketedDDP
+
+
+class DenoisingDecoder(nn.Sequential):
+    """Top-level embedding -> image patches (README.md:78-81)."""
+
+    def __init__(self, dim: int, image_size: int, patch_size: int):
+        side = image_size // patch_size
+        super().__init__(
+            nn.Linear(dim, patch_size ** 2 * 3),
+            Rearrange("b (h w) (p1 p2 c) -> b c (h p1) (w p2)",
+                      h=side, w=side, p1=patch_size, p2=patch_size),
+        )
+
+
+class DenoisingTrainer:
+    def __init__(self, model, *, lr=3e-4, noise_std=1.0, decode_step=7,
+                 distributed=False, bucket_bytes=16 << 20, log_path=None):
+        self.model = model
+        self.dim = model.dim
+        self.decode_step = decode_step
+        self.noise_std = noise_std
+        p = next(model.parameters())
+        self.decoder = DenoisingDecoder(
+            model.dim, model.image_size, model.patch_size
+        ).to(p.device, p.dtype)
+        self.distributed = distributed and dist.is_initialized()
+        self.ddp_model = self.ddp_dec = None
+        if self.distributed:
+            self.ddp_model = BucketedDDP(model, bucket_bytes)
+            self.ddp_dec = BucketedDDP(self.decoder, bucket_bytes)
+        self.opt = torch.optim.AdamW(
+            list(model.parameters()) + list(self.decoder.parameters()),
+            lr=lr, foreach=True)
+        self.step_idx = 0
+        self.log_path = log_path
+
+    def step(self, img: torch.Tensor, iters: int | None = None) -> float:
+        iters = iters if iters is not None else 2 * self.model.levels
+        t = min(self.decode_step, iters)
+        self.opt.zero_grad(set_to_none=True)
+        noised = img + torch.randn_like(img) * self.noise_std
+        all_levels = self.model(noised, iters=iters, return_all=True)
+        top = all_levels[t, :, :, -1]
+        recon = self.decoder(top)
+        loss = F.mse_loss(recon.float(), img.float())
+        loss.backward()
+        if self.distributed:
+            self.ddp_model.finalize()
+            self.ddp_dec.finalize()
+        self.opt.step()
+        self.step_idx += 1
+        return loss.item()
+
+    def log(self, **metrics):
+        if self.log_path:
+            with open(self.log_path, "a") as f:
+                f.write(json.dumps({"step": self.step_idx,
+                                    "time": time.time(), **metrics}) + "\n")
+
+    # -------------------- checkpoint / resume --------------------
+
+    def save_checkpoint(self, path: str):
+        if self.distributed and dist.get_rank() != 0:
+            return
+        tmp = path + ".tmp"
+        torch.save({
+            "model": self.model.state_dict(),
+            "decoder": self.decoder.state_dict(),
+            "optimizer": self.opt.state_dict(),
+            "step": self.step_idx,
+            "rng": torch.get_rng_state(),
+            "cuda_rng": (torch.cuda.get_rng_state()
+                         if torch.cuda.is_available() else None),
+        }, tmp)
+        os.replace(tmp, path)
+
+    def load_checkpoint(self, path: str):
+        ckpt = torch.load(path, map_location="cpu", weights_only=False)
+        self.model.load_state_dict(ckpt["model"])
+        self.decoder.load_state_dict(ckpt["decoder"])
+        self.opt.load_state_dict(ckpt["optimizer"])
+        self.step_idx = ckpt["step"]
+        torch.set_rng_state(ckpt["rng"])
+        if ckpt.get("cuda_rng") is not None and torch.cuda.is_available():
+            torch.cuda.set_rng_state(ckpt["cuda_rng"])
+        return self

@@ -1,0 +1,192 @@
+"""GPU parity tests: the CDNA4 HIP engine vs plain-PyTorch fp32 references.
+
+Every op test compares the bf16 HIP kernel against fp32 torch math computed
+from the SAME bf16 inputs, so tolerances only cover bf16 rounding inside the
+kernels.
+"""
+
+import os
+
+import pytest
+import torch
+
+from glom_pytorch_amd import Glom
+
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU"),
+]
+
+CFG = dict(dim=64, levels=3, image_size=32, patch_size=8)
+DEV = "cuda:0"
+
+
+def _rel_err(a, b):
+    a, b = a.float(), b.float()
+    return ((a - b).norm() / b.norm().clamp_min(1e-12)).item()
+
+
+def _models(**kw):
+    cfg = {**CFG, **kw}
+    torch.manual_seed(0)
+    m32 = Glom(**cfg).to(DEV)
+    m32.force_eager = True
+    mbf = Glom(**cfg).to(DEV)
+    mbf.load_state_dict(m32.state_dict())
+    mbf = mbf.to(torch.bfloat16)
+    return m32, mbf
+
+
+def test_extension_is_in_tree():
+    from glom_pytorch_amd.ops import _load_extension
+    ext = _load_extension()
+    import glom_pytorch_amd
+    pkg = os.path.dirname(os.path.dirname(glom_pytorch_amd.__file__))
+    assert ext.__file__.startswith(pkg), ext.__file__
+
+
+@pytest.mark.parametrize("kw", [
+    {},
+    dict(consensus_self=True),
+    dict(local_consensus_radius=2),
+])
+def test_forward_parity(kw):
+    m32, mbf = _models(**kw)
+    img = torch.randn(2, 3, 32, 32, device=DEV)
+    ref = m32(img, iters=3)
+    out = mbf(img.to(torch.bfloat16), iters=3)
+    assert out.dtype == torch.bfloat16
+    err = _rel_err(out, ref)
+    assert err < 2e-2, err
+
+
+def test_forward_parity_return_all_and_state():
+    m32, mbf = _models()
+    img = torch.randn(2, 3, 32, 32, device=DEV)
+    ref = m32(img, iters=4, return_all=True)
+    out = mbf(img.to(torch.bfloat16), iters=4, return_all=True)
+    assert out.shape == ref.shape
+    assert _rel_err(out, ref) < 2e-2
+    # stateful continuation
+    ref2 = m32(img, iters=2, levels=ref[-1])
+    out2 = mbf(img.to(torch.bfloat16), iters=2, levels=out[-1])
+    assert _rel_err(out2, ref2) < 3e-2
+
+
+def test_forward_parity_ragged_patches():
+    # N=9 patches (3x3 grid): exercises GEMM M/N/K tails and softmax tails
+    m32, mbf = _models(image_size=24, patch_size=8)
+    img = torch.randn(2, 3, 24, 24, device=DEV)
+    ref = m32(img, iters=3)
+    out = mbf(img.to(torch.bfloat16), iters=3)
+    assert _rel_err(out, ref) < 2e-2
+
+
+def test_backward_parity():
+    m32, mbf = _models()
+    img = torch.randn(2, 3, 32, 32, device=DEV)
+    for model, x in ((m32, img), (mbf, img.to(torch.bfloat16))):
+        out = model(x, iters=3, return_all=True)
+        loss = out[2, :, :, -1].float().pow(2).mean()
+        model.zero_grad()
+        loss.backward()
+    for (n32, p32), (nbf, pbf) in zip(m32.named_parameters(),
+                                      mbf.named_parameters()):
+        assert n32 == nbf
+        g32, gbf = p32.grad.float(), pbf.grad.float()
+        cos = torch.nn.functional.cosine_similarity(
+            g32.flatten(), gbf.flatten(), dim=0).item()
+        assert cos > 0.99, (n32, cos)
+        rel = _rel_err(gbf, g32)
+        assert rel < 0.1, (n32, rel)
+
+
+def test_grouped_ff_op():
+    from glom_pytorch_amd.ops.functional import GroupedFFFn
+    torch.manual_seed(1)
+    B, N, L, d = 2, 16, 3, 64
+    m4 = 4 * d
+    bf = torch.bfloat16
+    tokens = torch.randn(B, N, d, device=DEV, dtype=bf)
+    levels = torch.randn(B, N, L, d, device=DEV, dtype=bf)
+    pos = torch.randn(N, d, device=DEV, dtype=bf)
+    w1 = torch.randn(L * m4, d, device=DEV, dtype=bf) * 0.05
+    b1 = torch.randn(L * m4, device=DEV, dtype=bf)
+    w2 = torch.randn(L * d, m4, device=DEV, dtype=bf) * 0.05
+    b2 = torch.randn(L * d, device=DEV, dtype=bf)
+
+    out = GroupedFFFn.apply(tokens, levels, None, w1, b1, w2, b2, 0)
+    # fp32 reference from the same bf16 inputs
+    xs = [tokens.float()] + [levels[..., g, :].float() for g in range(L - 1)]
+    for g in range(L):
+        h = xs[g] @ w1.float()[g * m4:(g + 1) * m4].t() + b1.float()[g * m4:(g + 1) * m4]
+        y = torch.nn.functional.gelu(h) @ w2.float()[g * d:(g + 1) * d].t() \
+            + b2.float()[g * d:(g + 1) * d]
+        err = _rel_err(out[..., g, :], y)
+        assert err < 1e-2, (g, err)
+
+    # top-down mode with fused pos add (groups L-1)
+    w1t, b1t = w1[:(L - 1) * m4], b1[:(L - 1) * m4]
+    w2t, b2t = w2[:(L - 1) * d], b2[:(L - 1) * d]
+    out_td = GroupedFFFn.apply(None, levels, pos, w1t, b1t, w2t, b2t, 1)
+    for g in range(L - 1):
+        x = (levels[..., g + 1, :] + pos.view(1, N, d)).float()
+        h = x @ w1t.float()[g * m4:(g + 1) * m4].t() + b1t.float()[g * m4:(g + 1) * m4]
+        y = torch.nn.functional.gelu(h) @ w2t.float()[g * d:(g + 1) * d].t() \
+            + b2t.float()[g * d:(g + 1) * d]
+        err = _rel_err(out_td[..., g, :], y)
+        assert err < 1e-2, (g, err)
+
+
+def test_consensus_op():
+    from glom_pytorch_amd.ops.functional import ConsensusFn
+    import math
+    import torch.nn.functional as F
+    torch.manual_seed(2)
+    B, N, L, d = 2, 16, 3, 64
+    levels = torch.randn(B, N, L, d, device=DEV, dtype=torch.bfloat16)
+    for attend_self in (True, False):
+        out = ConsensusFn.apply(levels, attend_self, None)
+        lv = levels.float()
+        q, k = lv, F.normalize(lv, dim=-1)
+        sim = torch.einsum("bild,bjld->blij", q, k) / math.sqrt(d)
+        if not attend_self:
+            eye = torch.eye(N, device=DEV, dtype=torch.bool)
+            sim = sim.masked_fill(eye.view(1, 1, N, N), -5e-4)
+        ref = torch.einsum("blij,bjld->bild", sim.softmax(-1), lv)
+        err = _rel_err(out, ref)
+        assert err < 1e-2, (attend_self, err)
+
+
+def test_consensus_op_backward():
+    from glom_pytorch_amd.ops.functional import ConsensusFn
+    import math
+    import torch.nn.functional as F
+    torch.manual_seed(3)
+    B, N, L, d = 2, 16, 2, 64
+    lv_bf = torch.randn(B, N, L, d, device=DEV, dtype=torch.bfloat16,
+                        requires_grad=True)
+    out = ConsensusFn.apply(lv_bf, False, None)
+    g = torch.randn_like(out)
+    (dlev,) = torch.autograd.grad(out, lv_bf, g)
+
+    lv = lv_bf.detach().float().requires_grad_(True)
+    q, k = lv, F.normalize(lv, dim=-1)
+    sim = torch.einsum("bild,bjld->blij", q, k) / math.sqrt(d)
+    eye = torch.eye(N, device=DEV, dtype=torch.bool)
+    sim = sim.masked_fill(eye.view(1, 1, N, N), -5e-4)
+    ref = torch.einsum("blij,bjld->bild", sim.softmax(-1), lv)
+    (dref,) = torch.autograd.grad(ref, lv, g.float())
+    err = _rel_err(dlev, dref)
+    assert err < 3e-2, err
+
+
+def test_flagship_shapes_native():
+    torch.manual_seed(0)
+    m = Glom(dim=512, levels=6, image_size=224, patch_size=14)
+    m = m.to(DEV, torch.bfloat16)
+    img = torch.randn(2, 3, 224, 224, device=DEV, dtype=torch.bfloat16)
+    with torch.no_grad():
+        out = m(img, iters=12, return_all=True)
+    assert out.shape == (13, 2, 256, 6, 512)
+    assert torch.isfinite(out.float()).all()
