@@ -87,18 +87,22 @@ def test_backward_parity():
     img = torch.randn(2, 3, 32, 32, device=DEV)
     for model, x in ((m32, img), (mbf, img.to(torch.bfloat16))):
         out = model(x, iters=3, return_all=True)
-        loss = out[2, :, :, -1].float().pow(2).mean()
+        # touch ALL levels at the end plus the top level mid-trajectory so
+        # every parameter (incl. the patch embedding) receives gradient
+        loss = (out[-1].float().pow(2).mean()
+                + out[2, :, :, -1].float().pow(2).mean())
         model.zero_grad()
         loss.backward()
     for (n32, p32), (nbf, pbf) in zip(m32.named_parameters(),
                                       mbf.named_parameters()):
         assert n32 == nbf
         g32, gbf = p32.grad.float(), pbf.grad.float()
+        assert g32.norm() > 0, (n32, "fp32 grad unexpectedly zero")
         cos = torch.nn.functional.cosine_similarity(
             g32.flatten(), gbf.flatten(), dim=0).item()
         assert cos > 0.99, (n32, cos)
         rel = _rel_err(gbf, g32)
-        assert rel < 0.1, (n32, rel)
+        assert rel < 0.15, (n32, rel)
 
 
 def test_grouped_ff_op():

@@ -164,3 +164,18 @@ def test_bottom_up_sees_tokens_as_level_minus_1():
     b2 = m.bottom_up.net[3].bias[:d]
     ref0 = F.gelu(tokens @ w1.t() + b1) @ w2.t() + b2
     assert torch.allclose(bu[..., 0, :], ref0, atol=1e-5)
+
+
+def test_token_grad_depth():
+    """A top-level-only loss at trajectory index t reaches the patch tokens
+    only when t >= levels: level grads shift down exactly one level per
+    backward iteration through the bottom-up chain. (Found while validating
+    the HIP engine: cos(0,0)=0 looked like a kernel bug but is semantics.)"""
+    m = Glom(**SMALL)  # levels = 3
+    img = torch.randn(1, 3, 32, 32)
+    for t, expect_nonzero in ((2, False), (3, True)):
+        m.zero_grad()
+        out = m(img, iters=4, return_all=True)
+        out[t, :, :, -1].pow(2).mean().backward()
+        g = m.image_to_tokens[1].weight.grad
+        assert (g.norm().item() > 0) == expect_nonzero, (t, g.norm().item())
