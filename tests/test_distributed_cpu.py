@@ -22,7 +22,7 @@ def _grads_single(img_all):
     return {n: p.grad.clone() for n, p in model.named_parameters()}
 
 
-def _worker(rank, world, port, img_all, q):
+def _worker(rank, world, port, img_all, out_path):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -36,27 +36,28 @@ def _worker(rank, world, port, img_all, q):
     loss.backward()
     ddp.finalize()
     if rank == 0:
-        q.put({n: p.grad.clone() for n, p in model.named_parameters()})
+        torch.save({n: p.grad.clone() for n, p in model.named_parameters()},
+                   out_path)
     dist.destroy_process_group()
 
 
 @pytest.mark.timeout(300)
-def test_ddp_grad_parity_vs_single_process():
+def test_ddp_grad_parity_vs_single_process(tmp_path):
     torch.manual_seed(42)
     img_all = torch.randn(4, 3, 32, 32)
     ref = _grads_single(img_all)
 
+    out_path = str(tmp_path / "grads.pt")
     ctx = mp.get_context("spawn")
-    q = ctx.Queue()
     port = 29537
-    procs = [ctx.Process(target=_worker, args=(r, 2, port, img_all, q))
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, img_all, out_path))
              for r in range(2)]
     for p in procs:
         p.start()
-    got = q.get(timeout=240)
     for p in procs:
-        p.join(timeout=60)
+        p.join(timeout=240)
         assert p.exitcode == 0
+    got = torch.load(out_path, weights_only=False)
 
     # mean over 2 equal shards of the per-shard mean losses == big-batch mean
     for n, g in ref.items():
