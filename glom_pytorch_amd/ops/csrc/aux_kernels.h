@@ -1,6 +1,8 @@
 #pragma once
 #include <hip/hip_runtime.h>
 
+void launch_add_pos(const void* levels, const void* pos, void* out,
+                    long total, int N, int L, int d, hipStream_t s);
 void launch_rnorm(const void* levels, float* out, int B, int N, int L, int d,
                   hipStream_t s);
 void launch_softmax_fwd(const void* scores, void* probs, const bool* mask,

@@ -190,9 +190,39 @@ __global__ __launch_bounds__(NTHREADS) void k_mix_bwd(
     }
 }
 
+// td_in[b,n,g,:] = levels[b,n,g+1,:] + pos[n,:]  for g in 0..L-2
+// (the top-down input of reference glom_pytorch.py:136, materialized once
+// per iteration so the GEMM can stream it with plain 16B loads)
+__global__ __launch_bounds__(NTHREADS) void k_add_pos(
+        const ushort_t* __restrict__ levels, const ushort_t* __restrict__ pos,
+        ushort_t* __restrict__ out, long total, int N, int L, int d) {
+    long i8 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
+    if (i8 >= total) return;
+    int G = L - 1;
+    int q = i8 % d;
+    int g = (i8 / d) % G;
+    long bn = i8 / ((long)d * G);
+    long n = bn % N;
+    const ushort_t* lv = levels + (bn * L + g + 1) * (long)d + q;
+    const ushort_t* pp = pos + n * (long)d + q;
+    union { uint4v v; ushort_t u[8]; } a, b;
+    a.v = *(const uint4v*)lv;
+    b.v = *(const uint4v*)pp;
+#pragma unroll
+    for (int e = 0; e < 8; e++) a.u[e] = f2bf(bf2f(a.u[e]) + bf2f(b.u[e]));
+    *(uint4v*)(out + i8) = a.v;
+}
+
 // ---------------- host launchers ----------------
 
 static inline long cdiv(long a, long b) { return (a + b - 1) / b; }
+
+void launch_add_pos(const void* levels, const void* pos, void* out,
+                    long total, int N, int L, int d, hipStream_t s) {
+    hipLaunchKernelGGL(k_add_pos, dim3(cdiv(total / 8, NTHREADS)),
+                       dim3(NTHREADS), 0, s, (const ushort_t*)levels,
+                       (const ushort_t*)pos, (ushort_t*)out, total, N, L, d);
+}
 
 void launch_rnorm(const void* levels, float* out, int B, int N, int L, int d,
                   hipStream_t s) {

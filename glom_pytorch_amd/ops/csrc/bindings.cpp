@@ -31,6 +31,50 @@ void check_launch() {
     TORCH_CHECK(e == hipSuccess, "HIP launch failed: ", hipGetErrorString(e));
 }
 
+// Central dispatcher: picks the tuned full-tile kernels (glds NT / repack
+// TN) when the host-verified preconditions hold, applies split-K when the
+// natural grid underfills the 256 CUs (the weight-grad TN GEMMs: tiny
+// output, K = B*N tokens; partial f32 slabs summed by gemm_finish), and
+// otherwise falls back to the generic predicated kernel.
+// `lds_ok` asserts every A/B row stride (incl. table entries) is %8 == 0.
+void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
+              bool lds_ok) {
+    const bool no_xform = !(p.A.flags & (OP_GELU | OP_POS))
+                          && !(p.B.flags & (OP_GELU | OP_POS));
+    const bool nt_fast = p.layout == LAYOUT_NT && lds_ok && no_xform
+                         && p.M % 128 == 0 && p.N % 128 == 0
+                         && p.K % 64 == 0 && p.K >= 64;
+    const bool tn_fast = p.layout == LAYOUT_TN && lds_ok && no_xform
+                         && p.M % 128 == 0 && p.N % 128 == 0 && p.K % 8 == 0;
+
+    p.splitk = 1;
+    p.ws = nullptr;
+    torch::Tensor ws;
+    long blocks = ((p.N + 127) / 128) * ((p.M + 127) / 128) * p.nproblems;
+    if (p.layout == LAYOUT_TN && p.epilogue == EPI_NONE && !p.has_bias
+        && !p.has_colscale && blocks < 1024 && p.K >= 4096) {
+        long sk = std::min<long>(16, std::max<long>(1, 2048 / blocks));
+        sk = std::min<long>(sk, p.K / 64);
+        if (sk > 1) {
+            ws = torch::empty({sk * p.nproblems * (long)p.M * p.N},
+                              opts.dtype(at::kFloat));
+            p.splitk = (int)sk;
+            p.ws = ws.data_ptr<float>();
+        }
+    }
+    if (nt_fast)
+        launch_gemm_nt_fast(p, s);
+    else if (tn_fast)
+        launch_gemm_tn_fast(p, s);
+    else
+        launch_gemm(p, s);
+    check_launch();
+    if (p.splitk > 1) {
+        launch_gemm_finish(p, s);
+        check_launch();
+    }
+}
+
 GemmParams base_params(int64_t M, int64_t N, int64_t K, int layout,
                        int64_t nproblems, int64_t nInner, float alpha) {
     GemmParams p;
@@ -65,10 +109,24 @@ std::vector<torch::Tensor> grouped_ff_fwd(
 
     auto opts = levels.options();
     auto Hpre = torch::empty({G, M, m4}, opts);
+    auto Hact = torch::empty({G, M, m4}, opts);
     auto Y = torch::empty({B, N, G, d}, opts);
     hipStream_t s = cur_stream();
 
-    // up-projection: Hpre_g = x_g @ W1_g^T + b1_g
+    torch::Tensor td_in;
+    if (mode == 1) {
+        // materialize td_in = levels[..., 1:, :] + pos once, so the GEMM
+        // streams it with plain 16B loads / glds
+        TORCH_CHECK(pos_opt.has_value(), "top-down needs pos");
+        auto pos = pos_opt.value();
+        CHECK_IN(pos);
+        td_in = torch::empty({B, N, G, d}, opts);
+        launch_add_pos(levels.data_ptr(), pos.data_ptr(), td_in.data_ptr(),
+                       td_in.numel(), (int)N, (int)L, (int)d, s);
+        check_launch();
+    }
+
+    // up-projection: Hpre_g = x_g @ W1_g^T + b1_g ; Hact_g = gelu(Hpre_g)
     {
         GemmParams p = base_params(M, m4, d, LAYOUT_NT, G, G, 1.0f);
         if (mode == 0) {
@@ -83,40 +141,35 @@ std::vector<torch::Tensor> grouped_ff_fwd(
                 p.Atabld[g] = L * d;
             }
         } else {
-            TORCH_CHECK(pos_opt.has_value(), "top-down needs pos");
-            auto pos = pos_opt.value();
-            CHECK_IN(pos);
-            p.A.base = (const char*)levels.data_ptr() + d * 2;
-            p.A.sin = d; p.A.sout = 0; p.A.ld = L * d;
-            p.A.flags = OP_POS;
-            p.pos = pos.data_ptr();
-            p.pos_ld = d;
-            p.npatch = (int)N;
+            p.A.base = td_in.data_ptr();
+            p.A.sin = d; p.A.ld = G * d;
         }
         p.B.base = w1.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
         p.Cbase = Hpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
         p.bias_base = b1.data_ptr(); p.bias_sin = m4; p.has_bias = 1;
-        launch_gemm(p, s); check_launch();
+        p.epilogue = EPI_GELU_PAIR;
+        p.out2 = Hact.data_ptr(); p.out2_sin = M * m4; p.out2_ld = m4;
+        run_gemm(p, s, opts, true);
     }
-    // down-projection: Y_g = gelu(Hpre_g) @ W2_g^T + b2_g
+    // down-projection: Y_g = Hact_g @ W2_g^T + b2_g
     {
         GemmParams p = base_params(M, d, m4, LAYOUT_NT, G, G, 1.0f);
-        p.A.base = Hpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
-        p.A.flags = OP_GELU;
+        p.A.base = Hact.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
         p.B.base = w2.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
         p.Cbase = (char*)Y.data_ptr() ; p.Csin = d; p.Cld = G * d;
         p.bias_base = b2.data_ptr(); p.bias_sin = d; p.has_bias = 1;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, true);
     }
-    return {Y, Hpre};
+    return {Y, Hpre, Hact};
 }
 
 std::vector<torch::Tensor> grouped_ff_bwd(
         torch::Tensor dY, c10::optional<torch::Tensor> tokens_opt,
         torch::Tensor levels, c10::optional<torch::Tensor> pos_opt,
         torch::Tensor w1, torch::Tensor w2, torch::Tensor Hpre,
-        int64_t mode) {
+        torch::Tensor Hact, int64_t mode) {
     CHECK_IN(dY); CHECK_IN(levels); CHECK_IN(w1); CHECK_IN(w2); CHECK_IN(Hpre);
+    CHECK_IN(Hact);
     const int64_t B = levels.size(0), N = levels.size(1),
                   L = levels.size(2), d = levels.size(3);
     const int64_t G = (mode == 0) ? L : L - 1;
@@ -129,21 +182,35 @@ std::vector<torch::Tensor> grouped_ff_bwd(
     auto dLevels = torch::zeros({B, N, L, d}, opts);
     torch::Tensor dTokens;
 
-    // dHpre = (dY_g @ W2_g) * gelu'(Hpre_g)
+    // per-group weight transposes turn the NN data grads into NT GEMMs
+    auto w1t = w1.view({G, m4, d}).transpose(1, 2).contiguous();  // (G,d,m4)
+    auto w2t = w2.view({G, d, m4}).transpose(1, 2).contiguous();  // (G,m4,d)
+
+    torch::Tensor td_in;
+    if (mode == 1) {
+        auto pos = pos_opt.value();
+        CHECK_IN(pos);
+        td_in = torch::empty({B, N, G, d}, opts);
+        launch_add_pos(levels.data_ptr(), pos.data_ptr(), td_in.data_ptr(),
+                       td_in.numel(), (int)N, (int)L, (int)d, s);
+        check_launch();
+    }
+
+    // dHpre = (dY_g @ W2_g) * gelu'(Hpre_g)  -- NT with W2^T
     {
-        GemmParams p = base_params(M, m4, d, LAYOUT_NN, G, G, 1.0f);
+        GemmParams p = base_params(M, m4, d, LAYOUT_NT, G, G, 1.0f);
         p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
-        p.B.base = w2.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
+        p.B.base = w2t.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
         p.Cbase = dHpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
         p.epilogue = EPI_GELUGRAD;
         p.aux_base = Hpre.data_ptr(); p.aux_sin = M * m4; p.aux_ld = m4;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, true);
     }
-    // dX_g = dHpre_g @ W1_g, scattered into the level slices (or tokens)
+    // dX_g = dHpre_g @ W1_g -- NT with W1^T, scattered into level slices
     {
-        GemmParams p = base_params(M, d, m4, LAYOUT_NN, G, G, 1.0f);
+        GemmParams p = base_params(M, d, m4, LAYOUT_NT, G, G, 1.0f);
         p.A.base = dHpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
-        p.B.base = w1.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
+        p.B.base = w1t.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
         if (mode == 0) {
             dTokens = torch::empty({B, N, d}, opts);
             p.Cflags = OP_TABLE;
@@ -157,7 +224,7 @@ std::vector<torch::Tensor> grouped_ff_bwd(
             p.Cbase = (char*)dLevels.data_ptr() + d * 2;
             p.Csin = d; p.Cld = L * d;
         }
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, true);
     }
     // dW1_g[h, j] = sum_m dHpre_g[m, h] * x_g[m, j]
     auto dW1 = torch::empty({G * m4, d}, opts);
@@ -175,27 +242,20 @@ std::vector<torch::Tensor> grouped_ff_bwd(
                 p.Btabld[g] = L * d;
             }
         } else {
-            auto pos = pos_opt.value();
-            CHECK_IN(pos);
-            p.B.base = (const char*)levels.data_ptr() + d * 2;
-            p.B.sin = d; p.B.ld = L * d;
-            p.B.flags = OP_POS;
-            p.pos = pos.data_ptr();
-            p.pos_ld = d;
-            p.npatch = (int)N;
+            p.B.base = td_in.data_ptr();
+            p.B.sin = d; p.B.ld = G * d;
         }
         p.Cbase = dW1.data_ptr(); p.Csin = m4 * d; p.Cld = d;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, true);
     }
-    // dW2_g[o, h] = sum_m dY_g[m, o] * gelu(Hpre_g[m, h])
+    // dW2_g[o, h] = sum_m dY_g[m, o] * Hact_g[m, h]
     auto dW2 = torch::empty({G * d, m4}, opts);
     {
         GemmParams p = base_params(d, m4, M, LAYOUT_TN, G, G, 1.0f);
         p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
-        p.B.base = Hpre.data_ptr(); p.B.sin = M * m4; p.B.ld = m4;
-        p.B.flags = OP_GELU;
+        p.B.base = Hact.data_ptr(); p.B.sin = M * m4; p.B.ld = m4;
         p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, true);
     }
     auto dB1 = dHpre.sum(1).flatten();                       // (G*m4)
     auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();       // (G*d)
@@ -223,6 +283,7 @@ std::vector<torch::Tensor> consensus_fwd(
         mask = m.data_ptr<bool>();
     }
 
+    const bool lds_ok = (N % 8 == 0) && (d % 8 == 0);
     auto rnorm = torch::empty({B, L, N}, opts.dtype(at::kFloat));
     launch_rnorm(levels.data_ptr(), rnorm.data_ptr<float>(),
                  (int)B, (int)N, (int)L, (int)d, s);
@@ -240,7 +301,7 @@ std::vector<torch::Tensor> consensus_fwd(
         p.Cld = N;
         p.colscale_base = rnorm.data_ptr<float>();
         p.cs_sin = N; p.cs_sout = L * N; p.has_colscale = 1;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, lds_ok);
     }
     // masked softmax in place
     launch_softmax_fwd(probs.data_ptr(), probs.data_ptr(), mask, (int)P,
@@ -256,7 +317,7 @@ std::vector<torch::Tensor> consensus_fwd(
         p.B.ld = L * d;
         p.Cbase = out.data_ptr(); p.Csin = d; p.Csout = N * L * d;
         p.Cld = L * d;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, lds_ok);
     }
     return {out, probs, rnorm};
 }
@@ -269,6 +330,7 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
     const int64_t B = levels.size(0), N = levels.size(1),
                   L = levels.size(2), d = levels.size(3);
     const int64_t P = B * L;
+    const bool lds_ok = (N % 8 == 0) && (d % 8 == 0);
     auto opts = levels.options();
     hipStream_t s = cur_stream();
     const bool* mask = nullptr;
@@ -284,7 +346,7 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
         p.B.ld = L * d;
         p.Cbase = dP.data_ptr(); p.Csin = N * N; p.Csout = L * N * N;
         p.Cld = N;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, lds_ok);
     }
     auto dS = torch::empty_like(dP);
     auto dSr = torch::empty_like(dP);
@@ -304,7 +366,7 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
         p.B.ld = L * d;
         p.Cbase = dv.data_ptr(); p.Csin = d; p.Csout = N * L * d;
         p.Cld = L * d;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, lds_ok);
     }
     // dq[i,:] = sum_j dSr[i,j] levels[j,:]
     auto dq = torch::empty({B, N, L, d}, opts);
@@ -316,7 +378,7 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
         p.B.ld = L * d;
         p.Cbase = dq.data_ptr(); p.Csin = d; p.Csout = N * L * d;
         p.Cld = L * d;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, lds_ok);
     }
     // dkhat[j,:] = sum_i dS[i,j] levels[i,:]   (layout (B,L,N,d))
     auto dkhat = torch::empty({B, L, N, d}, opts);
@@ -328,7 +390,7 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
         p.B.ld = L * d;
         p.Cbase = dkhat.data_ptr(); p.Csin = N * d; p.Csout = L * N * d;
         p.Cld = d;
-        launch_gemm(p, s); check_launch();
+        run_gemm(p, s, opts, lds_ok);
     }
     auto dLevels = torch::empty({B, N, L, d}, opts);
     launch_knorm_combine(dkhat.data_ptr(), levels.data_ptr(),

@@ -5,7 +5,7 @@
 
 enum GemmLayout { LAYOUT_NT = 0, LAYOUT_NN = 1, LAYOUT_TN = 2 };
 enum OpFlags { OP_GELU = 1, OP_POS = 2, OP_TABLE = 4 };
-enum GemmEpilogue { EPI_NONE = 0, EPI_GELUGRAD = 1 };
+enum GemmEpilogue { EPI_NONE = 0, EPI_GELUGRAD = 1, EPI_GELU_PAIR = 2 };
 
 #define GEMM_MAX_TABLE 16
 
@@ -37,9 +37,27 @@ struct GemmParams {
 
     const void* pos; long pos_ld;
 
+    // split-K (reduction split): when > 1, the kernel writes per-slice f32
+    // partials into ws[slice][problem][M][N] and gemm_finish sums them into
+    // C (only EPI_NONE, no bias/colscale). Used for the weight-grad GEMMs
+    // whose natural grids underfill the 256 CUs (K = B*N tokens is huge).
+    int splitk; float* ws;
+
+    // EPI_GELU_PAIR: C gets the pre-activation (acc*alpha + bias), out2
+    // gets gelu(pre-activation). Used by the FF up-projection so the
+    // down-projection streams plain activations.
+    void* out2; long out2_sin, out2_sout, out2_ld;
+
+    // host-verified: every A/B row stride is a multiple of 8 elements and
+    // tiles are full -> the glds/repack fast kernels may run
+    int fast_ok;
+
     const void* Atab[GEMM_MAX_TABLE]; long Atabld[GEMM_MAX_TABLE];
     const void* Btab[GEMM_MAX_TABLE]; long Btabld[GEMM_MAX_TABLE];
     void* Ctab[GEMM_MAX_TABLE]; long Ctabld[GEMM_MAX_TABLE];
 };
 
 void launch_gemm(const GemmParams& p, hipStream_t stream);
+void launch_gemm_finish(const GemmParams& p, hipStream_t stream);
+void launch_gemm_nt_fast(const GemmParams& p, hipStream_t stream);
+void launch_gemm_tn_fast(const GemmParams& p, hipStream_t stream);

@@ -24,6 +24,7 @@
 
 #include "common.h"
 #include "gemm.h"
+#include "gemm_device.h"
 
 #define BM 128
 #define BN 128
@@ -31,23 +32,11 @@
 #define BKP (BK + 8)
 #define NTHREADS 256
 
-__device__ __forceinline__ void resolve_ptr(const OpArg& a,
-                                            const void* const* tab,
-                                            const long* tabld, int pid,
-                                            int nInner, const ushort_t** ptr,
-                                            long* ld) {
-    if (a.flags & OP_TABLE) {
-        *ptr = (const ushort_t*)tab[pid];
-        *ld = tabld[pid];
-    } else {
-        *ptr = (const ushort_t*)a.base + (long)(pid % nInner) * a.sin
-               + (long)(pid / nInner) * a.sout;
-        *ld = a.ld;
-    }
-}
-
 // Stage a [R x BK] tile of a row-major (rows x cols, leading dim ld) source
 // into LDS dst[R][BKP]. Rows r0.., cols k0..; out-of-range -> 0.
+// Fast path: 16-byte vector load + 16-byte LDS store when the row is fully
+// in range and 16B-aligned (ld % 8 == 0); transforms (pos-add / GELU) are
+// applied in registers between load and store.
 template <int R>
 __device__ __forceinline__ void stage_normal(
         ushort_t* dst, const ushort_t* src, long ld, int r0, int k0,
@@ -55,6 +44,7 @@ __device__ __forceinline__ void stage_normal(
         int npatch) {
     constexpr int CH = R * (BK / 8);
     const bool xform = (flags & (OP_GELU | OP_POS)) != 0;
+    const bool aligned = (ld % 8 == 0) && (((uintptr_t)src & 15) == 0);
 #pragma unroll
     for (int c0 = 0; c0 < CH; c0 += NTHREADS) {
         int c = c0 + threadIdx.x;
@@ -62,19 +52,17 @@ __device__ __forceinline__ void stage_normal(
         int row = c / (BK / 8);
         int kp = c % (BK / 8);
         int gr = r0 + row, gk = k0 + kp * 8;
-        ushort_t u[8];
-        if (gr < maxR && gk + 7 < maxK) {
-            const ushort_t* s = src + (long)gr * ld + gk;
-#pragma unroll
-            for (int e = 0; e < 8; e++) u[e] = s[e];
+        union { uint4v v; ushort_t u[8]; } t;
+        bool full = gr < maxR && gk + 7 < maxK;
+        if (full && aligned) {
+            t.v = *(const uint4v*)(src + (long)gr * ld + gk);
         } else if (gr < maxR) {
 #pragma unroll
             for (int e = 0; e < 8; e++)
-                u[e] = (gk + e < maxK) ? src[(long)gr * ld + gk + e]
-                                       : (ushort_t)0;
+                t.u[e] = (gk + e < maxK) ? src[(long)gr * ld + gk + e]
+                                         : (ushort_t)0;
         } else {
-#pragma unroll
-            for (int e = 0; e < 8; e++) u[e] = 0;
+            t.v = 0;
         }
         if (xform && gr < maxR) {
             const ushort_t* prow =
@@ -83,16 +71,14 @@ __device__ __forceinline__ void stage_normal(
 #pragma unroll
             for (int e = 0; e < 8; e++) {
                 if (gk + e < maxK) {
-                    float v = bf2f(u[e]);
+                    float v = bf2f(t.u[e]);
                     if (flags & OP_POS) v += bf2f(prow[e]);
                     if (flags & OP_GELU) v = gelu_f(v);
-                    u[e] = f2bf(v);
+                    t.u[e] = f2bf(v);
                 }
             }
         }
-        ushort_t* d = dst + row * BKP + kp * 8;
-#pragma unroll
-        for (int e = 0; e < 8; e++) d[e] = u[e];
+        *(uint4v*)(dst + row * BKP + kp * 8) = t.v;
     }
 }
 
@@ -106,6 +92,7 @@ __device__ __forceinline__ void stage_transposed(
         int npatch) {
     constexpr int CH = BK * (R / 8);
     const bool xform = (flags & (OP_GELU | OP_POS)) != 0;
+    const bool aligned = (ld % 8 == 0) && (((uintptr_t)src & 15) == 0);
 #pragma unroll
     for (int cc0 = 0; cc0 < CH; cc0 += NTHREADS) {
         int c = cc0 + threadIdx.x;
@@ -113,19 +100,17 @@ __device__ __forceinline__ void stage_transposed(
         int trow = c / (R / 8);
         int cp = c % (R / 8);
         int gt = t0 + trow, gc = c0 + cp * 8;
-        ushort_t u[8];
-        if (gt < maxT && gc + 7 < maxC) {
-            const ushort_t* s = src + (long)gt * ld + gc;
-#pragma unroll
-            for (int e = 0; e < 8; e++) u[e] = s[e];
+        union { uint4v v; ushort_t u[8]; } tt;
+        ushort_t* u = tt.u;
+        if (gt < maxT && gc + 7 < maxC && aligned) {
+            tt.v = *(const uint4v*)(src + (long)gt * ld + gc);
         } else if (gt < maxT) {
 #pragma unroll
             for (int e = 0; e < 8; e++)
                 u[e] = (gc + e < maxC) ? src[(long)gt * ld + gc + e]
                                        : (ushort_t)0;
         } else {
-#pragma unroll
-            for (int e = 0; e < 8; e++) u[e] = 0;
+            tt.v = 0;
         }
         if (xform && gt < maxT) {
             const ushort_t* prow =
@@ -150,15 +135,23 @@ __global__ __launch_bounds__(NTHREADS) void gemm_kernel(GemmParams p) {
     __shared__ ushort_t As[BM * BKP];
     __shared__ ushort_t Bs[BN * BKP];
 
-    const int pid = blockIdx.z;
+    int pid = blockIdx.z;
+    int slice = 0, k_begin = 0, k_end = p.K;
+    if (p.splitk > 1) {
+        pid = blockIdx.z % p.nproblems;
+        slice = blockIdx.z / p.nproblems;
+        int per = ((p.K + BK - 1) / BK + p.splitk - 1) / p.splitk * BK;
+        k_begin = slice * per;
+        k_end = min(p.K, k_begin + per);
+    }
     const int m0 = blockIdx.y * BM;
     const int n0 = blockIdx.x * BN;
 
     const ushort_t* Ap;
     const ushort_t* Bp;
     long lda, ldb;
-    resolve_ptr(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
-    resolve_ptr(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
 
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
@@ -171,7 +164,7 @@ __global__ __launch_bounds__(NTHREADS) void gemm_kernel(GemmParams p) {
 
     const ushort_t* pos = (const ushort_t*)p.pos;
 
-    for (int k0 = 0; k0 < p.K; k0 += BK) {
+    for (int k0 = k_begin; k0 < k_end; k0 += BK) {
         if (p.layout == LAYOUT_TN)
             stage_transposed<BM>(As, Ap, lda, k0, m0, p.K, p.M, p.A.flags,
                                  pos, p.pos_ld, p.npatch);
@@ -202,7 +195,39 @@ __global__ __launch_bounds__(NTHREADS) void gemm_kernel(GemmParams p) {
         __syncthreads();
     }
 
-    // ---- epilogue ----
+    // ---- split-K epilogue: raw f32 partials into this slice's slab ----
+    if (p.splitk > 1) {
+        float* ws = p.ws + ((long)slice * p.nproblems + pid) * p.M * p.N;
+#pragma unroll
+        for (int i16 = 0; i16 < 4; i16++) {
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int j = n0 + wn + j16 * 16 + lrow;
+                if (j >= p.N) continue;
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int i = m0 + wm + i16 * 16 + kq * 4 + r;
+                    if (i >= p.M) continue;
+                    ws[(long)i * p.N + j] = acc[i16][j16][r];
+                }
+            }
+        }
+        return;
+    }
+
+    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+}
+
+// Sum the split-K slices and write bf16 C (alpha applied once here).
+__global__ __launch_bounds__(NTHREADS) void gemm_finish_kernel(GemmParams p) {
+    long mn = (long)p.M * p.N;
+    long idx = (long)blockIdx.x * NTHREADS + threadIdx.x;
+    if (idx >= (long)p.nproblems * mn) return;
+    int pid = idx / mn;
+    long rem = idx % mn;
+    float s = 0.f;
+    for (int sl = 0; sl < p.splitk; sl++)
+        s += p.ws[((long)sl * p.nproblems + pid) * mn + rem];
     ushort_t* Cp;
     long ldc;
     {
@@ -210,47 +235,23 @@ __global__ __launch_bounds__(NTHREADS) void gemm_kernel(GemmParams p) {
         OpArg ca;
         ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
         ca.flags = p.Cflags;
-        resolve_ptr(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
-                    &tmp, &ldc);
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
         Cp = (ushort_t*)tmp;
     }
-    const ushort_t* biasp = nullptr;
-    if (p.has_bias)
-        biasp = (const ushort_t*)p.bias_base + (long)(pid % p.nInner) * p.bias_sin
-                + (long)(pid / p.nInner) * p.bias_sout;
-    const float* csp = nullptr;
-    if (p.has_colscale)
-        csp = (const float*)p.colscale_base + (long)(pid % p.nInner) * p.cs_sin
-              + (long)(pid / p.nInner) * p.cs_sout;
-    const ushort_t* auxp = nullptr;
-    if (p.epilogue == EPI_GELUGRAD)
-        auxp = (const ushort_t*)p.aux_base + (long)(pid % p.nInner) * p.aux_sin
-               + (long)(pid / p.nInner) * p.aux_sout;
-
-#pragma unroll
-    for (int i16 = 0; i16 < 4; i16++) {
-#pragma unroll
-        for (int j16 = 0; j16 < 4; j16++) {
-            int j = n0 + wn + j16 * 16 + lrow;
-            if (j >= p.N) continue;
-            float cs = csp ? csp[j] : 1.0f;
-            float bv = biasp ? bf2f(biasp[j]) : 0.0f;
-#pragma unroll
-            for (int r = 0; r < 4; r++) {
-                int i = m0 + wm + i16 * 16 + kq * 4 + r;
-                if (i >= p.M) continue;
-                float v = acc[i16][j16][r] * p.alpha;
-                v *= cs;
-                if (p.epilogue == EPI_GELUGRAD)
-                    v *= gelu_grad_f(bf2f(auxp[(long)i * p.aux_ld + j]));
-                v += bv;
-                Cp[(long)i * ldc + j] = f2bf(v);
-            }
-        }
-    }
+    long i = rem / p.N, j = rem % p.N;
+    Cp[i * ldc + j] = f2bf(s * p.alpha);
 }
 
 void launch_gemm(const GemmParams& p, hipStream_t stream) {
-    dim3 grid((p.N + BN - 1) / BN, (p.M + BM - 1) / BM, p.nproblems);
+    int sk = p.splitk > 1 ? p.splitk : 1;
+    dim3 grid((p.N + BN - 1) / BN, (p.M + BM - 1) / BM, p.nproblems * sk);
     hipLaunchKernelGGL(gemm_kernel, grid, dim3(NTHREADS), 0, stream, p);
+}
+
+void launch_gemm_finish(const GemmParams& p, hipStream_t stream) {
+    long total = (long)p.nproblems * p.M * p.N;
+    hipLaunchKernelGGL(gemm_finish_kernel,
+                       dim3((total + NTHREADS - 1) / NTHREADS),
+                       dim3(NTHREADS), 0, stream, p);
 }

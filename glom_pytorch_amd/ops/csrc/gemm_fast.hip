@@ -1,0 +1,244 @@
+// Tuned full-tile GEMM kernels for the GLOM hot path (gfx950).
+//
+// gemm_nt_fast: 128x128x64 tiles, 4 waves x (64x64), double-buffered LDS
+// filled by direct global->LDS DMA (`global_load_lds` width 16) with the
+// T2 XOR swizzle applied on the per-lane SOURCE address and again on the
+// ds_read_b128 fragment reads (linear lane-order LDS destination, as
+// rule 21 requires). One barrier per K-tile; the barrier's implicit
+// vmcnt(0) drains the in-flight DMA of the next tile.
+//
+// gemm_tn_fast: both operands have the reduction (token) index as the
+// SOURCE row, so tiles are transpose-staged: each thread loads an 8x8
+// block with 16B vector loads, repacks to column vectors in registers,
+// and writes 8 x ds_write_b128 into the swizzled [out-dim][k] image —
+// replacing the v1 per-element LDS scatter (8x fewer, 8x wider writes).
+//
+// Preconditions (host-checked, fast_ok): M%128==0, N%128==0, K%64==0
+// (TN: K%8==0 with row predicates), all row strides %8==0, no on-load
+// transforms. Everything else falls back to the generic gemm_kernel.
+
+#include "common.h"
+#include "gemm.h"
+#include "gemm_device.h"
+
+#define BM 128
+#define BN 128
+#define FBK 64
+#define NTHREADS 256
+
+// LDS bank swizzle: 16B-granule index in a 128B row is XORed with
+// swz_row(row); includes row bit 3 so writes/reads 8 rows apart do not
+// collide on the same 16B slot of the 256B bank row.
+__device__ __forceinline__ int swz_row(int row) {
+    return (row ^ (row >> 3)) & 7;
+}
+
+// Issue this wave's share of one operand tile (128 rows x 64 cols bf16 =
+// 16 KB = 16 DMA chunks of 1 KB; 4 chunks per wave) into a LINEAR LDS
+// image, pre-swizzling the source address so a swizzled ds_read_b128 is
+// conflict-free (byte_in_row ^= (row&7)<<4).
+__device__ __forceinline__ void stage_glds(
+        ushort_t* lds, const ushort_t* src, long ld, int r0, int k0,
+        int wid, int lane) {
+#pragma unroll
+    for (int c = 0; c < 4; c++) {
+        int chunk = wid * 4 + c;
+        int row = chunk * 8 + (lane >> 3);
+        int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+        const ushort_t* gaddr = src + (long)(r0 + row) * ld + k0 + swz8;
+        ushort_t* laddr = lds + chunk * 512;  // wave-uniform chunk base
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gaddr,
+            (__attribute__((address_space(3))) unsigned int*)laddr, 16, 0, 0);
+    }
+}
+
+__global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
+    __shared__ ushort_t As[2][BM * FBK];
+    __shared__ ushort_t Bs[2][BN * FBK];
+
+    const int pid = blockIdx.z;
+    const int m0 = blockIdx.y * BM;
+    const int n0 = blockIdx.x * BN;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 1) * 64;
+    const int wn = (wid & 1) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    const int nk = p.K / FBK;
+    int cur = 0;
+    stage_glds(As[0], Ap, lda, m0, 0, wid, lane);
+    stage_glds(Bs[0], Bp, ldb, n0, 0, wid, lane);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    for (int kt = 0; kt < nk; kt++) {
+        if (kt + 1 < nk) {
+            stage_glds(As[cur ^ 1], Ap, lda, m0, (kt + 1) * FBK, wid, lane);
+            stage_glds(Bs[cur ^ 1], Bp, ldb, n0, (kt + 1) * FBK, wid, lane);
+        }
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&As[cur][row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bs[cur][row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        __syncthreads();   // drains the in-flight DMA (vmcnt0) + buffer reuse
+        cur ^= 1;
+    }
+    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+}
+
+// Transpose-stage one operand tile: source rows = reduction slice
+// [t0, t0+64) (row stride ld), cols = output-dim slice [c0, c0+128).
+// 128 threads per operand; thread (mb, cb) loads an 8(m) x 8(c) block and
+// writes 8 column-vectors of 8 bf16 via ds_write_b128 into the swizzled
+// [col][m] image (row length 64 elements = 128 B).
+__device__ __forceinline__ void stage_repack(
+        ushort_t* lds, const ushort_t* src, long ld, int t0, int c0,
+        int kend, int tid128) {
+    int cb = tid128 & 15;   // 16 col-blocks of 8
+    int mb = tid128 >> 4;   // 8 m-blocks of 8
+    union { uint4v v; ushort_t u[8]; } rowv[8];
+#pragma unroll
+    for (int r = 0; r < 8; r++) {
+        int m = t0 + mb * 8 + r;
+        if (m < kend)
+            rowv[r].v = *(const uint4v*)(src + (long)m * ld + c0 + cb * 8);
+        else
+            rowv[r].v = 0;
+    }
+#pragma unroll
+    for (int e = 0; e < 8; e++) {
+        union { uint4v v; ushort_t u[8]; } col;
+#pragma unroll
+        for (int r = 0; r < 8; r++) col.u[r] = rowv[r].u[e];
+        int row = cb * 8 + e;                       // output-dim index
+        int off = (mb * 8) ^ (swz_row(row) << 3);   // swizzled m offset
+        *(uint4v*)&lds[row * FBK + off] = col.v;
+    }
+}
+
+__global__ __launch_bounds__(NTHREADS) void gemm_tn_fast_kernel(GemmParams p) {
+    __shared__ ushort_t As[BM * FBK];
+    __shared__ ushort_t Bs[BN * FBK];
+
+    int pid = blockIdx.z;
+    int slice = 0, k_begin = 0, k_end = p.K;
+    if (p.splitk > 1) {
+        pid = blockIdx.z % p.nproblems;
+        slice = blockIdx.z / p.nproblems;
+        int per = ((p.K + FBK - 1) / FBK + p.splitk - 1) / p.splitk * FBK;
+        k_begin = slice * per;
+        k_end = min(p.K, k_begin + per);
+    }
+    const int m0 = blockIdx.y * BM;
+    const int n0 = blockIdx.x * BN;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 1) * 64;
+    const int wn = (wid & 1) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    for (int k0 = k_begin; k0 < k_end; k0 += FBK) {
+        if (threadIdx.x < 128)
+            stage_repack(As, Ap, lda, k0, m0, k_end, threadIdx.x);
+        else
+            stage_repack(Bs, Bp, ldb, k0, n0, k_end, threadIdx.x - 128);
+        __syncthreads();
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&As[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bs[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+
+    if (p.splitk > 1) {
+        float* ws = p.ws + ((long)slice * p.nproblems + pid) * p.M * p.N;
+#pragma unroll
+        for (int i16 = 0; i16 < 4; i16++)
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int j = n0 + wn + j16 * 16 + lrow;
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int i = m0 + wm + i16 * 16 + kq * 4 + r;
+                    ws[(long)i * p.N + j] = acc[i16][j16][r];
+                }
+            }
+        return;
+    }
+    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+}
+
+void launch_gemm_nt_fast(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN, p.M / BM, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast_kernel, grid, dim3(NTHREADS), 0, stream,
+                       p);
+}
+
+void launch_gemm_tn_fast(const GemmParams& p, hipStream_t stream) {
+    int sk = p.splitk > 1 ? p.splitk : 1;
+    dim3 grid(p.N / BN, p.M / BM, p.nproblems * sk);
+    hipLaunchKernelGGL(gemm_tn_fast_kernel, grid, dim3(NTHREADS), 0, stream,
+                       p);
+}

@@ -26,18 +26,19 @@ class GroupedFFFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, tokens, levels, pos, w1, b1, w2, b2, mode):
         ext = _load_extension()
-        Y, Hpre = ext.grouped_ff_fwd(tokens, levels, pos, w1, b1, w2, b2,
-                                     mode)
-        ctx.save_for_backward(tokens, levels, pos, w1, w2, Hpre)
+        Y, Hpre, Hact = ext.grouped_ff_fwd(tokens, levels, pos, w1, b1, w2,
+                                           b2, mode)
+        ctx.save_for_backward(tokens, levels, pos, w1, w2, Hpre, Hact)
         ctx.mode = mode
         return Y
 
     @staticmethod
     def backward(ctx, dY):
         ext = _load_extension()
-        tokens, levels, pos, w1, w2, Hpre = ctx.saved_tensors
+        tokens, levels, pos, w1, w2, Hpre, Hact = ctx.saved_tensors
         dTokens, dLevels, dW1, dB1, dW2, dB2 = ext.grouped_ff_bwd(
-            dY.contiguous(), tokens, levels, pos, w1, w2, Hpre, ctx.mode)
+            dY.contiguous(), tokens, levels, pos, w1, w2, Hpre, Hact,
+            ctx.mode)
         dPos = None
         if ctx.mode == 1 and pos is not None and ctx.needs_input_grad[2]:
             # pos was added to every top-down group input; its grad is the

@@ -18,6 +18,7 @@ ext = CUDAExtension(
     sources=[
         "glom_pytorch_amd/ops/csrc/bindings.cpp",
         "glom_pytorch_amd/ops/csrc/gemm.hip",
+        "glom_pytorch_amd/ops/csrc/gemm_fast.hip",
         "glom_pytorch_amd/ops/csrc/aux_kernels.hip",
     ],
     extra_compile_args={

@@ -194,3 +194,30 @@ def test_flagship_shapes_native():
         out = m(img, iters=12, return_all=True)
     assert out.shape == (13, 2, 256, 6, 512)
     assert torch.isfinite(out.float()).all()
+
+
+def test_parity_fast_tile_shapes():
+    """Shapes that are full 128-multiples exercise the tuned glds NT and
+    repack TN kernels (small-dim tests above cover the generic fallback)."""
+    cfg = dict(dim=128, levels=3, image_size=128, patch_size=8)  # N=256
+    torch.manual_seed(0)
+    m32 = Glom(**cfg).to(DEV)
+    m32.force_eager = True
+    mbf = Glom(**cfg).to(DEV)
+    mbf.load_state_dict(m32.state_dict())
+    mbf = mbf.to(torch.bfloat16)
+    img = torch.randn(1, 3, 128, 128, device=DEV)
+    for model, x in ((m32, img), (mbf, img.to(torch.bfloat16))):
+        out = model(x, iters=3, return_all=True)
+        loss = out[-1].float().pow(2).mean()
+        model.zero_grad()
+        loss.backward()
+    out32 = m32(img, iters=3)
+    outbf = mbf(img.to(torch.bfloat16), iters=3)
+    assert _rel_err(outbf, out32) < 2e-2
+    for (n32, p32), (nbf, pbf) in zip(m32.named_parameters(),
+                                      mbf.named_parameters()):
+        g32, gbf = p32.grad.float(), pbf.grad.float()
+        cos = torch.nn.functional.cosine_similarity(
+            g32.flatten(), gbf.flatten(), dim=0).item()
+        assert cos > 0.99, (n32, cos)
