@@ -58,8 +58,16 @@ __global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
     __shared__ ushort_t Bs[2][BN * FBK];
 
     const int pid = blockIdx.z;
-    const int m0 = blockIdx.y * BM;
-    const int n0 = blockIdx.x * BN;
+    // XCD-aware block remap (T1): give each XCD die a contiguous chunk of
+    // this problem's tile grid so neighbouring tiles hit the same L2.
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int m0 = (bid / gridDim.x) * BM;
+    const int n0 = (bid % gridDim.x) * BN;
 
     const ushort_t* Ap;
     const ushort_t* Bp;
@@ -161,8 +169,16 @@ __global__ __launch_bounds__(NTHREADS) void gemm_tn_fast_kernel(GemmParams p) {
         k_begin = slice * per;
         k_end = min(p.K, k_begin + per);
     }
-    const int m0 = blockIdx.y * BM;
-    const int n0 = blockIdx.x * BN;
+    // XCD-aware block remap (T1): give each XCD die a contiguous chunk of
+    // this problem's tile grid so neighbouring tiles hit the same L2.
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int m0 = (bid / gridDim.x) * BM;
+    const int n0 = (bid % gridDim.x) * BN;
 
     const ushort_t* Ap;
     const ushort_t* Bp;
