@@ -34,7 +34,7 @@ __device__ __forceinline__ ushort_t f2bf(float f) {
 // erff, which dominated the GEMM epilogues at 200M gelu evals per launch.
 __device__ __forceinline__ float fast_erf(float x) {
     float ax = fabsf(x);
-    float t = __frcp_rn(1.0f + 0.3275911f * ax);
+    float t = __builtin_amdgcn_rcpf(1.0f + 0.3275911f * ax);
     float poly = t * (0.254829592f
                       + t * (-0.284496736f
                              + t * (1.421413741f

@@ -56,25 +56,25 @@ __device__ __forceinline__ void gemm_epilogue(
 #pragma unroll
     for (int i16 = 0; i16 < 4; i16++) {
 #pragma unroll
-        for (int j16 = 0; j16 < 4; j16++) {
-            int j = n0 + wn + j16 * 16 + lrow;
-            if (j >= p.N) continue;
-            float cs = csp ? csp[j] : 1.0f;
-            float bv = biasp ? bf2f(biasp[j]) : 0.0f;
+        for (int r = 0; r < 4; r++) {
+            int i = m0 + wm + i16 * 16 + kq * 4 + r;
+            if (i >= p.M) continue;
+            ushort_t* crow = Cp + (long)i * ldc;
+            ushort_t* o2row =
+                out2p ? out2p + (long)i * p.out2_ld : nullptr;
+            const ushort_t* auxrow =
+                auxp ? auxp + (long)i * p.aux_ld : nullptr;
 #pragma unroll
-            for (int r = 0; r < 4; r++) {
-                int i = m0 + wm + i16 * 16 + kq * 4 + r;
-                if (i >= p.M) continue;
+            for (int j16 = 0; j16 < 4; j16++) {
+                int j = n0 + wn + j16 * 16 + lrow;
+                if (j >= p.N) continue;
                 float v = acc[i16][j16][r] * p.alpha;
-                v *= cs;
-                if (p.epilogue == EPI_GELUGRAD)
-                    v *= gelu_grad_f(bf2f(auxp[(long)i * p.aux_ld + j]));
-                v += bv;
-                Cp[(long)i * ldc + j] = f2bf(v);
-                if (p.epilogue == EPI_GELU_PAIR)
-                    out2p[(long)i * p.out2_ld + j] = f2bf(gelu_f(v));
+                if (csp) v *= csp[j];
+                if (auxrow) v *= gelu_grad_f(bf2f(auxrow[j]));
+                if (biasp) v += bf2f(biasp[j]);
+                crow[j] = f2bf(v);
+                if (o2row) o2row[j] = f2bf(gelu_f(v));
             }
         }
     }
 }
-

@@ -54,8 +54,12 @@ __device__ __forceinline__ void stage_glds(
 }
 
 __global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
-    __shared__ ushort_t As[2][BM * FBK];
-    __shared__ ushort_t Bs[2][BN * FBK];
+    // ONE shared array: a second __shared__ object makes hipcc emit
+    // s_waitcnt vmcnt(0) before the first ds_read of every k-step of a
+    // glds pipeline, draining the prefetch (guide §5 ".s-level traps" (a)).
+    __shared__ ushort_t smem[4 * BM * FBK];
+    ushort_t* As0 = smem;
+    ushort_t* Bs0 = smem + 2 * BM * FBK;
 
     const int pid = blockIdx.z;
     // XCD-aware block remap (T1): give each XCD die a contiguous chunk of
@@ -86,15 +90,17 @@ __global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
 
     const int nk = p.K / FBK;
     int cur = 0;
-    stage_glds(As[0], Ap, lda, m0, 0, wid, lane);
-    stage_glds(Bs[0], Bp, ldb, n0, 0, wid, lane);
+    stage_glds(As0, Ap, lda, m0, 0, wid, lane);
+    stage_glds(Bs0, Bp, ldb, n0, 0, wid, lane);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
 
     for (int kt = 0; kt < nk; kt++) {
         if (kt + 1 < nk) {
-            stage_glds(As[cur ^ 1], Ap, lda, m0, (kt + 1) * FBK, wid, lane);
-            stage_glds(Bs[cur ^ 1], Bp, ldb, n0, (kt + 1) * FBK, wid, lane);
+            stage_glds(As0 + (cur ^ 1) * BM * FBK, Ap, lda, m0,
+                       (kt + 1) * FBK, wid, lane);
+            stage_glds(Bs0 + (cur ^ 1) * BN * FBK, Bp, ldb, n0,
+                       (kt + 1) * FBK, wid, lane);
         }
         short8 af[2][4], bfr[2][4];
 #pragma unroll
@@ -103,13 +109,13 @@ __global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
             for (int i = 0; i < 4; i++) {
                 int row = wm + i * 16 + lrow;
                 int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                af[s][i] = *(const short8*)&As[cur][row * FBK + off];
+                af[s][i] = *(const short8*)&As0[cur * BM * FBK + row * FBK + off];
             }
 #pragma unroll
             for (int j = 0; j < 4; j++) {
                 int row = wn + j * 16 + lrow;
                 int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                bfr[s][j] = *(const short8*)&Bs[cur][row * FBK + off];
+                bfr[s][j] = *(const short8*)&Bs0[cur * BN * FBK + row * FBK + off];
             }
         }
 #pragma unroll
