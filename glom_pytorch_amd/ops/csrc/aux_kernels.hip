@@ -32,9 +32,21 @@ __global__ __launch_bounds__(NTHREADS) void k_rnorm(
     long b = row / ((long)N * L);
     const ushort_t* x = levels + row * d;
     float ss = 0.f;
-    for (int q = lane; q < d; q += WAVE) {
-        float v = bf2f(x[q]);
-        ss += v * v;
+    if (d % 8 == 0) {
+        for (int q0 = lane * 8; q0 < d; q0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } t;
+            t.v = *(const uint4v*)(x + q0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) {
+                float v = bf2f(t.u[e]);
+                ss += v * v;
+            }
+        }
+    } else {
+        for (int q = lane; q < d; q += WAVE) {
+            float v = bf2f(x[q]);
+            ss += v * v;
+        }
     }
     ss = wave_reduce_sum(ss);
     if (lane == 0)
@@ -56,6 +68,52 @@ __global__ __launch_bounds__(NTHREADS) void k_softmax_fwd(
     const float SELF = bf2f(f2bf(-5e-4f));
     const float NEG = -3.3895314e38f;  // -finfo(bf16).max
 
+    if (N % 8 == 0) {
+        // vectorized: each lane owns contiguous 8-element chunks; the row
+        // (< 1024 values at the stretch config) is re-read from L1/L2
+        float mx = -INFINITY, sum = 0.f;
+#pragma unroll 1
+        for (int j0 = lane * 8; j0 < N; j0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } t;
+            t.v = *(const uint4v*)(s + j0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) {
+                float v = bf2f(t.u[e]);
+                if (self_mask && j0 + e == i) v = SELF;
+                if (m && m[j0 + e]) v = NEG;
+                mx = fmaxf(mx, v);
+            }
+        }
+        mx = wave_reduce_max(mx);
+        mx = __shfl(mx, 0, WAVE);
+#pragma unroll 1
+        for (int j0 = lane * 8; j0 < N; j0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } t, ov;
+            t.v = *(const uint4v*)(s + j0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) {
+                float v = bf2f(t.u[e]);
+                if (self_mask && j0 + e == i) v = SELF;
+                if (m && m[j0 + e]) v = NEG;
+                float ex = __expf(v - mx);
+                sum += ex;
+                ov.u[e] = f2bf(ex);
+            }
+            *(uint4v*)(o + j0) = ov.v;   // unnormalized, rescaled below
+        }
+        sum = wave_reduce_sum(sum);
+        sum = __shfl(sum, 0, WAVE);
+        float inv = 1.0f / sum;
+#pragma unroll 1
+        for (int j0 = lane * 8; j0 < N; j0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } t;
+            t.v = *(const uint4v*)(o + j0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) t.u[e] = f2bf(bf2f(t.u[e]) * inv);
+            *(uint4v*)(o + j0) = t.v;
+        }
+        return;
+    }
     float mx = -INFINITY;
     for (int j = lane; j < N; j += WAVE) {
         float v = bf2f(s[j]);
@@ -101,6 +159,35 @@ __global__ __launch_bounds__(NTHREADS) void k_softmax_bwd(
     const bool* m = nonlocal_mask ? nonlocal_mask + (long)i * N : nullptr;
 
     float t = 0.f;
+    if (N % 8 == 0) {
+#pragma unroll 1
+        for (int j0 = lane * 8; j0 < N; j0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } a, bb;
+            a.v = *(const uint4v*)(pr + j0);
+            bb.v = *(const uint4v*)(dpr + j0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) t += bf2f(a.u[e]) * bf2f(bb.u[e]);
+        }
+        t = wave_reduce_sum(t);
+        t = __shfl(t, 0, WAVE);
+#pragma unroll 1
+        for (int j0 = lane * 8; j0 < N; j0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } a, bb, o1, o2;
+            a.v = *(const uint4v*)(pr + j0);
+            bb.v = *(const uint4v*)(dpr + j0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) {
+                float v = alpha * bf2f(a.u[e]) * (bf2f(bb.u[e]) - t);
+                if (self_mask && j0 + e == i) v = 0.f;
+                if (m && m[j0 + e]) v = 0.f;
+                o1.u[e] = f2bf(v);
+                o2.u[e] = f2bf(v * rn[j0 + e]);
+            }
+            *(uint4v*)(dS + row * N + j0) = o1.v;
+            *(uint4v*)(dSr + row * N + j0) = o2.v;
+        }
+        return;
+    }
     for (int j = lane; j < N; j += WAVE) t += bf2f(pr[j]) * bf2f(dpr[j]);
     t = wave_reduce_sum(t);
     t = __shfl(t, 0, WAVE);
@@ -134,18 +221,46 @@ __global__ __launch_bounds__(NTHREADS) void k_knorm_combine(
     float r = rnorm[(b * L + l) * N + n];
     bool clamped = r >= 0.999e12f;
 
+    const bool vec = (d % 8 == 0);
     float dot = 0.f;
     if (!clamped) {
-        for (int q = lane; q < d; q += WAVE)
-            dot += bf2f(dkh[q]) * bf2f(x[q]);
+        if (vec) {
+            for (int q0 = lane * 8; q0 < d; q0 += WAVE * 8) {
+                union { uint4v v; ushort_t u[8]; } a, b;
+                a.v = *(const uint4v*)(dkh + q0);
+                b.v = *(const uint4v*)(x + q0);
+#pragma unroll
+                for (int e = 0; e < 8; e++)
+                    dot += bf2f(a.u[e]) * bf2f(b.u[e]);
+            }
+        } else {
+            for (int q = lane; q < d; q += WAVE)
+                dot += bf2f(dkh[q]) * bf2f(x[q]);
+        }
         dot = wave_reduce_sum(dot);
         dot = __shfl(dot, 0, WAVE);
     }
     float c = clamped ? 0.f : r * r * dot;
-    for (int q = lane; q < d; q += WAVE) {
-        float dk = r * (bf2f(dkh[q]) - bf2f(x[q]) * c);
-        out[row * d + q] =
-            f2bf(dk + bf2f(dv[row * d + q]) + bf2f(dq[row * d + q]));
+    if (vec) {
+        for (int q0 = lane * 8; q0 < d; q0 += WAVE * 8) {
+            union { uint4v v; ushort_t u[8]; } a, b, dvv, dqv, o;
+            a.v = *(const uint4v*)(dkh + q0);
+            b.v = *(const uint4v*)(x + q0);
+            dvv.v = *(const uint4v*)(dv + row * d + q0);
+            dqv.v = *(const uint4v*)(dq + row * d + q0);
+#pragma unroll
+            for (int e = 0; e < 8; e++) {
+                float dk = r * (bf2f(a.u[e]) - bf2f(b.u[e]) * c);
+                o.u[e] = f2bf(dk + bf2f(dvv.u[e]) + bf2f(dqv.u[e]));
+            }
+            *(uint4v*)(out + row * d + q0) = o.v;
+        }
+    } else {
+        for (int q = lane; q < d; q += WAVE) {
+            float dk = r * (bf2f(dkh[q]) - bf2f(x[q]) * c);
+            out[row * d + q] =
+                f2bf(dk + bf2f(dv[row * d + q]) + bf2f(dq[row * d + q]));
+        }
     }
 }
 
@@ -155,39 +270,44 @@ __global__ __launch_bounds__(NTHREADS) void k_mix_fwd(
         const ushort_t* __restrict__ prev, const ushort_t* __restrict__ bu,
         const ushort_t* __restrict__ td, const ushort_t* __restrict__ cons,
         ushort_t* __restrict__ out, long total, int L, int d) {
-    long idx = (long)blockIdx.x * NTHREADS + threadIdx.x;
+    long idx = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
     if (idx >= total) return;
     int q = idx % d;
     int l = (idx / d) % L;
     long bn = idx / ((long)d * L);
-    float v = bf2f(prev[idx]) + bf2f(bu[idx]) + bf2f(cons[idx]);
-    float c;
-    if (l < L - 1) {
-        v += bf2f(td[(bn * (L - 1) + l) * (long)d + q]);
-        c = 4.0f;
-    } else {
-        c = 3.0f;
+    union { uint4v v; ushort_t u[8]; } a, b, c4, t, o;
+    a.v = *(const uint4v*)(prev + idx);
+    b.v = *(const uint4v*)(bu + idx);
+    c4.v = *(const uint4v*)(cons + idx);
+    bool has_td = l < L - 1;
+    if (has_td) t.v = *(const uint4v*)(td + (bn * (L - 1) + l) * (long)d + q);
+    float c = has_td ? 4.0f : 3.0f;
+#pragma unroll
+    for (int e = 0; e < 8; e++) {
+        float v = bf2f(a.u[e]) + bf2f(b.u[e]) + bf2f(c4.u[e]);
+        if (has_td) v += bf2f(t.u[e]);
+        o.u[e] = f2bf(v / c);
     }
-    out[idx] = f2bf(v / c);
+    *(uint4v*)(out + idx) = o.v;
 }
 
 // dmix = dout / c_l (shared by prev/bu/cons); dtd = dout[..., :L-1, :] / 4
 __global__ __launch_bounds__(NTHREADS) void k_mix_bwd(
         const ushort_t* __restrict__ dout, ushort_t* __restrict__ dmix,
         ushort_t* __restrict__ dtd, long total, int L, int d) {
-    long idx = (long)blockIdx.x * NTHREADS + threadIdx.x;
+    long idx = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
     if (idx >= total) return;
     int q = idx % d;
     int l = (idx / d) % L;
     long bn = idx / ((long)d * L);
-    float g = bf2f(dout[idx]);
-    if (l < L - 1) {
-        float v = g / 4.0f;
-        dmix[idx] = f2bf(v);
-        dtd[(bn * (L - 1) + l) * (long)d + q] = f2bf(v);
-    } else {
-        dmix[idx] = f2bf(g / 3.0f);
-    }
+    union { uint4v v; ushort_t u[8]; } g, o;
+    g.v = *(const uint4v*)(dout + idx);
+    bool has_td = l < L - 1;
+    float c = has_td ? 4.0f : 3.0f;
+#pragma unroll
+    for (int e = 0; e < 8; e++) o.u[e] = f2bf(bf2f(g.u[e]) / c);
+    *(uint4v*)(dmix + idx) = o.v;
+    if (has_td) *(uint4v*)(dtd + (bn * (L - 1) + l) * (long)d + q) = o.v;
 }
 
 // td_in[b,n,g,:] = levels[b,n,g+1,:] + pos[n,:]  for g in 0..L-2
@@ -263,7 +383,7 @@ void launch_knorm_combine(const void* dkhat, const void* lev,
 void launch_mix_fwd(const void* prev, const void* bu, const void* td,
                     const void* cons, void* out, long total, int L, int d,
                     hipStream_t s) {
-    hipLaunchKernelGGL(k_mix_fwd, dim3(cdiv(total, NTHREADS)), dim3(NTHREADS),
+    hipLaunchKernelGGL(k_mix_fwd, dim3(cdiv(total / 8, NTHREADS)), dim3(NTHREADS),
                        0, s, (const ushort_t*)prev, (const ushort_t*)bu,
                        (const ushort_t*)td, (const ushort_t*)cons,
                        (ushort_t*)out, total, L, d);
@@ -271,7 +391,7 @@ void launch_mix_fwd(const void* prev, const void* bu, const void* td,
 
 void launch_mix_bwd(const void* dout, void* dmix, void* dtd, long total,
                     int L, int d, hipStream_t s) {
-    hipLaunchKernelGGL(k_mix_bwd, dim3(cdiv(total, NTHREADS)), dim3(NTHREADS),
+    hipLaunchKernelGGL(k_mix_bwd, dim3(cdiv(total / 8, NTHREADS)), dim3(NTHREADS),
                        0, s, (const ushort_t*)dout, (ushort_t*)dmix,
                        (ushort_t*)dtd, total, L, d);
 }

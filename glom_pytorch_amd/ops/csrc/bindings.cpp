@@ -46,6 +46,8 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
                          && p.K % 64 == 0 && p.K >= 64;
     const bool tn_fast = p.layout == LAYOUT_TN && lds_ok && no_xform
                          && p.M % 128 == 0 && p.N % 128 == 0 && p.K % 8 == 0;
+    const bool nn_fast = p.layout == LAYOUT_NN && lds_ok && no_xform
+                         && p.M % 128 == 0 && p.N % 128 == 0 && p.K % 8 == 0;
 
     p.splitk = 1;
     p.ws = nullptr;
@@ -66,6 +68,8 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         launch_gemm_nt_fast(p, s);
     else if (tn_fast)
         launch_gemm_tn_fast(p, s);
+    else if (nn_fast)
+        launch_gemm_nn_fast(p, s);
     else
         launch_gemm(p, s);
     check_launch();
@@ -408,6 +412,7 @@ torch::Tensor level_mix_fwd(torch::Tensor levels, torch::Tensor bu,
                             torch::Tensor td, torch::Tensor cons) {
     CHECK_IN(levels); CHECK_IN(bu); CHECK_IN(td); CHECK_IN(cons);
     const int64_t L = levels.size(2), d = levels.size(3);
+    TORCH_CHECK(d % 8 == 0, "dim must be a multiple of 8");
     auto out = torch::empty_like(levels);
     launch_mix_fwd(levels.data_ptr(), bu.data_ptr(), td.data_ptr(),
                    cons.data_ptr(), out.data_ptr(), levels.numel(), (int)L,
@@ -420,6 +425,7 @@ std::vector<torch::Tensor> level_mix_bwd(torch::Tensor dout) {
     CHECK_IN(dout);
     const int64_t B = dout.size(0), N = dout.size(1), L = dout.size(2),
                   d = dout.size(3);
+    TORCH_CHECK(d % 8 == 0, "dim must be a multiple of 8");
     auto dmix = torch::empty_like(dout);
     auto dtd = torch::empty({B, N, L - 1, d}, dout.options());
     launch_mix_bwd(dout.data_ptr(), dmix.data_ptr(), dtd.data_ptr(),

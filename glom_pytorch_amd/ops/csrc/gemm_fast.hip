@@ -62,16 +62,18 @@ __global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
     ushort_t* Bs0 = smem + 2 * BM * FBK;
 
     const int pid = blockIdx.z;
-    // XCD-aware block remap (T1): give each XCD die a contiguous chunk of
-    // this problem's tile grid so neighbouring tiles hit the same L2.
+    // XCD-aware block remap (T1), column-major: each XCD die owns a
+    // contiguous run of N-columns (all M-tiles of a few n-tiles), so the
+    // shared B panel of a column stays resident in that XCD's private L2
+    // while the M sweep streams A.
     int nwg = gridDim.x * gridDim.y;
     int bid = blockIdx.y * gridDim.x + blockIdx.x;
     {
         int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
         bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
     }
-    const int m0 = (bid / gridDim.x) * BM;
-    const int n0 = (bid % gridDim.x) * BN;
+    const int n0 = (bid / gridDim.y) * BN;   // column-major: n outer
+    const int m0 = (bid % gridDim.y) * BM;
 
     const ushort_t* Ap;
     const ushort_t* Bp;
@@ -175,16 +177,18 @@ __global__ __launch_bounds__(NTHREADS) void gemm_tn_fast_kernel(GemmParams p) {
         k_begin = slice * per;
         k_end = min(p.K, k_begin + per);
     }
-    // XCD-aware block remap (T1): give each XCD die a contiguous chunk of
-    // this problem's tile grid so neighbouring tiles hit the same L2.
+    // XCD-aware block remap (T1), column-major: each XCD die owns a
+    // contiguous run of N-columns (all M-tiles of a few n-tiles), so the
+    // shared B panel of a column stays resident in that XCD's private L2
+    // while the M sweep streams A.
     int nwg = gridDim.x * gridDim.y;
     int bid = blockIdx.y * gridDim.x + blockIdx.x;
     {
         int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
         bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
     }
-    const int m0 = (bid / gridDim.x) * BM;
-    const int n0 = (bid % gridDim.x) * BN;
+    const int n0 = (bid / gridDim.y) * BN;   // column-major: n outer
+    const int m0 = (bid % gridDim.y) * BM;
 
     const ushort_t* Ap;
     const ushort_t* Bp;
@@ -262,5 +266,102 @@ void launch_gemm_tn_fast(const GemmParams& p, hipStream_t stream) {
     int sk = p.splitk > 1 ? p.splitk : 1;
     dim3 grid(p.N / BN, p.M / BM, p.nproblems * sk);
     hipLaunchKernelGGL(gemm_tn_fast_kernel, grid, dim3(NTHREADS), 0, stream,
+                       p);
+}
+
+// ---------------------------------------------------------------- //
+// NN fast kernel (attention AV / dq): A row-major [M][K] staged as a
+// plain swizzled copy; B row-major [K][N] transpose-staged with the
+// 8x8 register repack. Single LDS buffer, two barriers per K-step.
+
+__device__ __forceinline__ void stage_copy128(
+        ushort_t* lds, const ushort_t* src, long ld, int r0, int k0,
+        int maxK, int tid128) {
+    // 128 threads stage a 128x64 tile: thread t handles rows t, t+... no:
+    // 1024 chunks of 8 elements / 128 threads = 8 chunks each
+#pragma unroll
+    for (int cc = 0; cc < 8; cc++) {
+        int c = tid128 * 8 + cc;
+        int row = c >> 3;
+        int g = c & 7;
+        int gk = k0 + g * 8;
+        union { uint4v v; ushort_t u[8]; } t;
+        if (gk + 7 < maxK)
+            t.v = *(const uint4v*)(src + (long)(r0 + row) * ld + gk);
+        else
+            t.v = 0;
+        int off = (g * 8) ^ (swz_row(row) << 3);
+        *(uint4v*)&lds[row * FBK + off] = t.v;
+    }
+}
+
+__global__ __launch_bounds__(NTHREADS) void gemm_nn_fast_kernel(GemmParams p) {
+    __shared__ ushort_t smem[2 * BM * FBK];
+    ushort_t* As = smem;
+    ushort_t* Bs = smem + BM * FBK;
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN;
+    const int m0 = (bid % gridDim.y) * BM;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 1) * 64;
+    const int wn = (wid & 1) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    for (int k0 = 0; k0 < p.K; k0 += FBK) {
+        if (threadIdx.x < 128)
+            stage_copy128(As, Ap, lda, m0, k0, p.K, threadIdx.x);
+        else
+            stage_repack(Bs, Bp, ldb, k0, n0, p.K, threadIdx.x - 128);
+        __syncthreads();
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&As[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bs[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+}
+
+void launch_gemm_nn_fast(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN, p.M / BM, p.nproblems);
+    hipLaunchKernelGGL(gemm_nn_fast_kernel, grid, dim3(NTHREADS), 0, stream,
                        p);
 }
