@@ -1,0 +1,78 @@
+"""hipGraph capture of the GLOM iteration loop (north star).
+
+The T-step hot loop (reference glom_pytorch.py:131-145) is launch-bound at
+small batch: every iteration issues ~10 kernels whose launch latency is pure
+overhead under `torch.no_grad()`. Everything inside the loop is
+iteration-invariant except `levels` (tokens / pos / masks are loop
+constants), which makes the whole forward cleanly capturable: we record one
+hipGraph per (batch, image size, iters, return_all, stateful) key and replay
+it as a single graph launch, copying the new image (and incoming levels for
+the stateful video path, README.md:92-112) into static buffers.
+
+torch.cuda.CUDAGraph on ROCm IS hipGraph capture; replay re-runs the exact
+recorded kernel sequence of the CDNA4 engine.
+"""
+
+from __future__ import annotations
+
+import torch
+
+
+class _GraphEntry:
+    def __init__(self, model, img, iters, levels, return_all):
+        from glom_pytorch_amd.ops.functional import glom_forward
+        self.static_img = img.clone()
+        self.static_levels = levels.clone() if levels is not None else None
+        with torch.no_grad():
+            # warmup on a side stream so allocator state is steady
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    glom_forward(model, self.static_img, iters,
+                                 levels=self.static_levels,
+                                 return_all=return_all)
+            torch.cuda.current_stream().wait_stream(s)
+
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self.static_out = glom_forward(model, self.static_img, iters,
+                                               levels=self.static_levels,
+                                               return_all=return_all)
+
+    def replay(self, img, levels):
+        self.static_img.copy_(img)
+        if self.static_levels is not None:
+            self.static_levels.copy_(levels)
+        self.graph.replay()
+        # clone: callers may feed the output back in (video path) or keep it
+        # across subsequent replays
+        return self.static_out.clone()
+
+
+class GraphCache:
+    """Per-model cache of captured forwards, keyed on everything that
+    changes the kernel sequence."""
+
+    def __init__(self, model):
+        self.model = model
+        self.entries: dict = {}
+
+    def run(self, img, iters, levels, return_all):
+        key = (tuple(img.shape), img.dtype, iters, return_all,
+               levels is not None)
+        entry = self.entries.get(key)
+        if entry is None:
+            entry = _GraphEntry(self.model, img, iters, levels, return_all)
+            self.entries[key] = entry
+        return entry.replay(img, levels)
+
+
+def graphs_usable(model, img, levels) -> bool:
+    if torch.is_grad_enabled():
+        return False
+    if not img.is_cuda:
+        return False
+    if levels is not None and levels.requires_grad:
+        return False
+    return True

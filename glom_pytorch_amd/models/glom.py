@@ -180,10 +180,28 @@ class Glom(nn.Module):
         iters = iters if iters is not None else 2 * self.levels
 
         if self._use_native(img):
+            cache = getattr(self, "_graph_cache", None)
+            if cache is not None:
+                from glom_pytorch_amd.graphs import graphs_usable
+                if graphs_usable(self, img, levels):
+                    return cache.run(img, iters, levels, return_all)
             from glom_pytorch_amd.ops import native_forward
             return native_forward(self, img, iters=iters, levels=levels,
                                   return_all=return_all)
         return self._eager_forward(img, iters, levels, return_all)
+
+    def enable_graphs(self):
+        """hipGraph-capture the T-step loop for inference: each distinct
+        (shape, iters, return_all, stateful) forward is recorded once and
+        replayed as a single graph launch. No-grad forwards only; training
+        and CPU paths are unaffected."""
+        from glom_pytorch_amd.graphs import GraphCache
+        self._graph_cache = GraphCache(self)
+        return self
+
+    def disable_graphs(self):
+        self._graph_cache = None
+        return self
 
     # ------------------------------------------------------------------ #
     # Eager (plain PyTorch) path — the semantic specification. CPU tests
