@@ -434,6 +434,58 @@ std::vector<torch::Tensor> level_mix_bwd(torch::Tensor dout) {
     return {dmix, dtd};
 }
 
+// Synthetic single-GEMM microbenchmark for kernel tuning (used by
+// scripts/gemmbench.py; not part of the model path).
+double bench_gemm(int64_t M, int64_t N, int64_t K, int64_t layout,
+                  int64_t nproblems, int64_t epilogue, int64_t reps) {
+    auto opts = torch::TensorOptions().dtype(at::kBFloat16)
+                    .device(at::kCUDA);
+    torch::Tensor A, Bt;
+    if (layout == LAYOUT_TN)
+        A = torch::randn({nproblems, K, M}, opts) * 0.05;
+    else
+        A = torch::randn({nproblems, M, K}, opts) * 0.05;
+    if (layout == LAYOUT_NT)
+        Bt = torch::randn({nproblems, N, K}, opts) * 0.05;
+    else
+        Bt = torch::randn({nproblems, K, N}, opts) * 0.05;
+    auto C = torch::empty({nproblems, M, N}, opts);
+    auto bias = torch::randn({nproblems, N}, opts);
+    auto aux = torch::randn({nproblems, M, N}, opts);
+    auto out2 = torch::empty({nproblems, M, N}, opts);
+    hipStream_t s = cur_stream();
+
+    auto run = [&]() {
+        GemmParams p = base_params(M, N, K, (int)layout, nproblems,
+                                   nproblems, 1.0f);
+        p.A.base = A.data_ptr();
+        p.A.sin = A.size(1) * A.size(2); p.A.ld = A.size(2);
+        p.B.base = Bt.data_ptr();
+        p.B.sin = Bt.size(1) * Bt.size(2); p.B.ld = Bt.size(2);
+        p.Cbase = C.data_ptr(); p.Csin = M * N; p.Cld = N;
+        p.epilogue = (int)epilogue;
+        if (epilogue == EPI_GELUGRAD) {
+            p.aux_base = aux.data_ptr(); p.aux_sin = M * N; p.aux_ld = N;
+        }
+        if (epilogue == EPI_GELU_PAIR) {
+            p.bias_base = bias.data_ptr(); p.bias_sin = N; p.has_bias = 1;
+            p.out2 = out2.data_ptr(); p.out2_sin = M * N; p.out2_ld = N;
+        }
+        run_gemm(p, s, opts, true);
+    };
+    for (int i = 0; i < 3; i++) run();
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0); hipEventCreate(&e1);
+    hipEventRecord(e0, s);
+    for (int64_t i = 0; i < reps; i++) run();
+    hipEventRecord(e1, s);
+    hipEventSynchronize(e1);
+    float ms = 0;
+    hipEventElapsedTime(&ms, e0, e1);
+    hipEventDestroy(e0); hipEventDestroy(e1);
+    return ms / reps;
+}
+
 std::string build_info() {
     return "glom_pytorch_amd HIP extension (gfx950, bf16 MFMA 16x16x32)";
 }
@@ -448,4 +500,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("level_mix_fwd", &level_mix_fwd, "level mix forward");
     m.def("level_mix_bwd", &level_mix_bwd, "level mix backward");
     m.def("build_info", &build_info);
+    m.def("bench_gemm", &bench_gemm, "raw GEMM microbench (tuning only)");
 }
