@@ -131,7 +131,7 @@ __global__ __launch_bounds__(NTHREADS) void gemm_nt_fast_kernel(GemmParams p) {
         __syncthreads();   // drains the in-flight DMA (vmcnt0) + buffer reuse
         cur ^= 1;
     }
-    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+    gemm_epilogue_lds(p, pid, m0, n0, wm, wn, lrow, kq, acc, smem);
 }
 
 // Transpose-stage one operand tile: source rows = reduction slice
@@ -165,8 +165,9 @@ __device__ __forceinline__ void stage_repack(
 }
 
 __global__ __launch_bounds__(NTHREADS) void gemm_tn_fast_kernel(GemmParams p) {
-    __shared__ ushort_t As[BM * FBK];
-    __shared__ ushort_t Bs[BN * FBK];
+    __shared__ ushort_t smem[128 * EPI_LDS_ROW];   // >= As+Bs (2*8192)
+    ushort_t* As = smem;
+    ushort_t* Bs = smem + BM * FBK;
 
     int pid = blockIdx.z;
     int slice = 0, k_begin = 0, k_end = p.K;
@@ -253,7 +254,8 @@ __global__ __launch_bounds__(NTHREADS) void gemm_tn_fast_kernel(GemmParams p) {
             }
         return;
     }
-    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+    __syncthreads();   // all waves done with As/Bs before epilogue staging
+    gemm_epilogue_lds(p, pid, m0, n0, wm, wn, lrow, kq, acc, smem);
 }
 
 void launch_gemm_nt_fast(const GemmParams& p, hipStream_t stream) {
@@ -296,7 +298,7 @@ __device__ __forceinline__ void stage_copy128(
 }
 
 __global__ __launch_bounds__(NTHREADS) void gemm_nn_fast_kernel(GemmParams p) {
-    __shared__ ushort_t smem[2 * BM * FBK];
+    __shared__ ushort_t smem[128 * EPI_LDS_ROW];   // >= As+Bs (2*8192)
     ushort_t* As = smem;
     ushort_t* Bs = smem + BM * FBK;
 
@@ -357,7 +359,8 @@ __global__ __launch_bounds__(NTHREADS) void gemm_nn_fast_kernel(GemmParams p) {
                         af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
         __syncthreads();
     }
-    gemm_epilogue(p, pid, m0, n0, wm, wn, lrow, kq, acc);
+    __syncthreads();
+    gemm_epilogue_lds(p, pid, m0, n0, wm, wn, lrow, kq, acc, smem);
 }
 
 void launch_gemm_nn_fast(const GemmParams& p, hipStream_t stream) {
