@@ -78,3 +78,95 @@ __device__ __forceinline__ void gemm_epilogue(
         }
     }
 }
+
+// Full-tile epilogue with LDS-staged stores: the MFMA D-fragment layout
+// (col = lane&15, rows scattered) makes direct global stores 2-byte
+// partial-line writes (read-modify-write amplified in L2). Instead the
+// 128x128 output tile is packed to bf16 in LDS (padded rows), then
+// streamed out as full 128B lines. EPI_GELU_PAIR computes gelu() during
+// the linear pass (on the rounded bf16 pre-activation, exactly what the
+// eager reference's gelu(conv_out_bf16) computes).
+// Requires: full tile (no predicates), scratch >= 128*136*2 bytes.
+#define EPI_LDS_ROW 136   // 128 cols + 8 pad elements (16B-aligned rows)
+
+__device__ __forceinline__ void gemm_epilogue_lds(
+        const GemmParams& p, int pid, int m0, int n0, int wm, int wn,
+        int lrow, int kq, const f32x4 acc[4][4], ushort_t* scratch) {
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    ushort_t* out2p = nullptr;
+    if (p.epilogue == EPI_GELU_PAIR)
+        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout;
+
+    // phase 1: acc -> bf16 into LDS [128][EPI_LDS_ROW], local coords
+#pragma unroll
+    for (int i16 = 0; i16 < 4; i16++) {
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            int li = wm + i16 * 16 + kq * 4 + r;           // 0..127
+            int gi = m0 + li;
+            const ushort_t* auxrow =
+                auxp ? auxp + (long)gi * p.aux_ld : nullptr;
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int lj = wn + j16 * 16 + lrow;             // 0..127
+                int gj = n0 + lj;
+                float v = acc[i16][j16][r] * p.alpha;
+                if (csp) v *= csp[gj];
+                if (auxrow) v *= gelu_grad_f(bf2f(auxrow[gj]));
+                if (biasp) v += bf2f(biasp[gj]);
+                scratch[li * EPI_LDS_ROW + lj] = f2bf(v);
+            }
+        }
+    }
+    __syncthreads();
+    // phase 2: linear sweep, 2 threads per row, full 16B chunks
+    {
+        int t = threadIdx.x;           // 256 threads, 128 rows x 2 halves
+        int li = t >> 1;
+        int half = (t & 1) * 64;
+        long gi = m0 + li;
+        ushort_t* crow = Cp + gi * ldc + n0 + half;
+        ushort_t* orow =
+            out2p ? out2p + gi * p.out2_ld + n0 + half : nullptr;
+        const ushort_t* srow = scratch + li * EPI_LDS_ROW + half;
+#pragma unroll
+        for (int c = 0; c < 8; c++) {
+            union { uint4v v; ushort_t u[8]; } x;
+            x.v = *(const uint4v*)(srow + c * 8);
+            *(uint4v*)(crow + c * 8) = x.v;
+            if (orow) {
+                union { uint4v v; ushort_t u[8]; } g;
+#pragma unroll
+                for (int e = 0; e < 8; e++)
+                    g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+                *(uint4v*)(orow + c * 8) = g.v;
+            }
+        }
+    }
+}
