@@ -64,7 +64,10 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
             p.ws = ws.data_ptr<float>();
         }
     }
-    if (nt_fast)
+    const bool nt2 = nt_fast && p.M % 256 == 0 && p.N % 256 == 0;
+    if (nt2)
+        launch_gemm_nt_fast2(p, s);
+    else if (nt_fast)
         launch_gemm_nt_fast(p, s);
     else if (tn_fast)
         launch_gemm_tn_fast(p, s);

@@ -368,3 +368,195 @@ void launch_gemm_nn_fast(const GemmParams& p, hipStream_t stream) {
     hipLaunchKernelGGL(gemm_nn_fast_kernel, grid, dim3(NTHREADS), 0, stream,
                        p);
 }
+
+// ---------------------------------------------------------------- //
+// 256x256-tile NT kernel: 8 waves (512 threads), per-wave 128x64 output,
+// BK=64 double-buffered glds staging (128 KiB LDS, 1 block/CU, 2 waves/
+// SIMD). 4x the work per block of the 128^2 kernel at 2x the arithmetic
+// intensity (128 flops/byte) — amortizes the per-block prologue/epilogue
+// that dominates the short-K (K=512) GLOM shapes.
+
+#define NT2 512
+#define BM2 256
+#define BN2 256
+
+// stage one 256-row x 64-col operand tile: 64 KiB = 32 DMA chunks, 4 per wave
+__device__ __forceinline__ void stage_glds256(
+        ushort_t* lds, const ushort_t* src, long ld, int r0, int k0,
+        int wid, int lane) {
+#pragma unroll
+    for (int c = 0; c < 4; c++) {
+        int chunk = wid * 4 + c;
+        int row = chunk * 8 + (lane >> 3);
+        int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+        const ushort_t* gaddr = src + (long)(r0 + row) * ld + k0 + swz8;
+        ushort_t* laddr = lds + chunk * 512;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gaddr,
+            (__attribute__((address_space(3))) unsigned int*)laddr, 16, 0, 0);
+    }
+}
+
+#define EPI2_ROW 264   // 256 cols + 8 pad
+
+__global__ __launch_bounds__(NT2) void gemm_nt_fast2_kernel(GemmParams p) {
+    __shared__ ushort_t smem[4 * BM2 * FBK];   // 128 KiB: 2 bufs x (A|B)
+    ushort_t* As0 = smem;
+    ushort_t* Bs0 = smem + 2 * BM2 * FBK;
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN2;
+    const int m0 = (bid % gridDim.y) * BM2;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 2) * 128;
+    const int wn = (wid & 3) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[8][4] = {};
+
+    const int nk = p.K / FBK;
+    int cur = 0;
+    stage_glds256(As0, Ap, lda, m0, 0, wid, lane);
+    stage_glds256(Bs0, Bp, ldb, n0, 0, wid, lane);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    for (int kt = 0; kt < nk; kt++) {
+        if (kt + 1 < nk) {
+            stage_glds256(As0 + (cur ^ 1) * BM2 * FBK, Ap, lda, m0,
+                          (kt + 1) * FBK, wid, lane);
+            stage_glds256(Bs0 + (cur ^ 1) * BM2 * FBK, Bp, ldb, n0,
+                          (kt + 1) * FBK, wid, lane);
+        }
+        const ushort_t* Ab = As0 + cur * BM2 * FBK;
+        const ushort_t* Bb = Bs0 + cur * BM2 * FBK;
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+            // B fragments stay live; A fragments stream one row-block at a
+            // time so the 128-reg accumulator + frags fit the 256-VGPR/wave
+            // budget of 2 waves/SIMD (8-wave block)
+            short8 bfr[4];
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[j] = *(const short8*)&Bb[row * FBK + off];
+            }
+#pragma unroll
+            for (int i = 0; i < 8; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                short8 af = *(const short8*)&Ab[row * FBK + off];
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af, bfr[j], acc[i][j], 0, 0, 0);
+            }
+        }
+        __syncthreads();
+        cur ^= 1;
+    }
+
+    // ---- epilogue: two half-tiles (rows 0-127, 128-255) via LDS ----
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    ushort_t* out2p = nullptr;
+    if (p.epilogue == EPI_GELU_PAIR)
+        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout;
+
+#pragma unroll
+    for (int half = 0; half < 2; half++) {
+        if (wm == half * 128) {
+#pragma unroll
+            for (int i16 = 0; i16 < 8; i16++) {
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int li = i16 * 16 + kq * 4 + r;        // 0..127
+                    long gi = m0 + half * 128 + li;
+                    const ushort_t* auxrow =
+                        auxp ? auxp + gi * p.aux_ld : nullptr;
+#pragma unroll
+                    for (int j16 = 0; j16 < 4; j16++) {
+                        int lj = wn + j16 * 16 + lrow;     // 0..255
+                        int gj = n0 + lj;
+                        float v = acc[i16][j16][r] * p.alpha;
+                        if (csp) v *= csp[gj];
+                        if (auxrow) v *= gelu_grad_f(bf2f(auxrow[gj]));
+                        if (biasp) v += bf2f(biasp[gj]);
+                        smem[li * EPI2_ROW + lj] = f2bf(v);
+                    }
+                }
+            }
+        }
+        __syncthreads();
+        {
+            int t = threadIdx.x;          // 512 threads: 128 rows x 4 qtrs
+            int li = t >> 2;
+            int qt = (t & 3) * 64;
+            long gi = m0 + half * 128 + li;
+            ushort_t* crow = Cp + gi * ldc + n0 + qt;
+            ushort_t* orow =
+                out2p ? out2p + gi * p.out2_ld + n0 + qt : nullptr;
+            const ushort_t* srow = smem + li * EPI2_ROW + qt;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+                *(uint4v*)(crow + c * 8) = x.v;
+                if (orow) {
+                    union { uint4v v; ushort_t u[8]; } g;
+#pragma unroll
+                    for (int e = 0; e < 8; e++)
+                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+                    *(uint4v*)(orow + c * 8) = g.v;
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+void launch_gemm_nt_fast2(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN2, p.M / BM2, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast2_kernel, grid, dim3(NT2), 0, stream, p);
+}
