@@ -64,7 +64,10 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
             p.ws = ws.data_ptr<float>();
         }
     }
-    const bool nt2 = nt_fast && p.M % 256 == 0 && p.N % 256 == 0;
+    // 256^2 tile only pays with an 8-phase schedule (guide tile-space
+    // table); with the 2-barrier loop it regressed, so it stays off until
+    // the phase-pipelined variant lands.
+    const bool nt2 = false;
     if (nt2)
         launch_gemm_nt_fast2(p, s);
     else if (nt_fast)
@@ -470,7 +473,7 @@ double bench_gemm(int64_t M, int64_t N, int64_t K, int64_t layout,
         if (epilogue == EPI_GELUGRAD) {
             p.aux_base = aux.data_ptr(); p.aux_sin = M * N; p.aux_ld = N;
         }
-        if (epilogue == EPI_GELU_PAIR) {
+        if (epilogue == EPI_GELU_PAIR || epilogue == 10) {
             p.bias_base = bias.data_ptr(); p.bias_sin = N; p.has_bias = 1;
             p.out2 = out2.data_ptr(); p.out2_sin = M * N; p.out2_ld = N;
         }

@@ -119,7 +119,7 @@ __device__ __forceinline__ void gemm_epilogue_lds(
                + (long)(pid % p.nInner) * p.aux_sin
                + (long)(pid / p.nInner) * p.aux_sout;
     ushort_t* out2p = nullptr;
-    if (p.epilogue == EPI_GELU_PAIR)
+    if (p.epilogue == EPI_GELU_PAIR || p.epilogue == 10)
         out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
                 + (long)(pid / p.nInner) * p.out2_sout;
 
@@ -156,15 +156,20 @@ __device__ __forceinline__ void gemm_epilogue_lds(
             out2p ? out2p + gi * p.out2_ld + n0 + half : nullptr;
         const ushort_t* srow = scratch + li * EPI_LDS_ROW + half;
 #pragma unroll
-        for (int c = 0; c < 8; c++) {
-            union { uint4v v; ushort_t u[8]; } x;
-            x.v = *(const uint4v*)(srow + c * 8);
-            *(uint4v*)(crow + c * 8) = x.v;
-            if (orow) {
-                union { uint4v v; ushort_t u[8]; } g;
+        for (int c = 0; c < 8; c++)
+            *(uint4v*)(crow + c * 8) = *(const uint4v*)(srow + c * 8);
+        if (orow) {
 #pragma unroll
-                for (int e = 0; e < 8; e++)
-                    g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x, g;
+                x.v = *(const uint4v*)(srow + c * 8);
+                if (p.epilogue == 10) {       // debug: pair without gelu
+                    g.v = x.v;
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; e++)
+                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+                }
                 *(uint4v*)(orow + c * 8) = g.v;
             }
         }
