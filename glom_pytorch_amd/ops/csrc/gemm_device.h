@@ -123,24 +123,46 @@ __device__ __forceinline__ void gemm_epilogue_lds(
         out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
                 + (long)(pid / p.nInner) * p.out2_sout;
 
-    // phase 1: acc -> bf16 into LDS [128][EPI_LDS_ROW], local coords
+    // phase 1: acc -> bf16 into LDS [128][EPI_LDS_ROW], local coords.
+    // Column factors are pre-loaded under ONE branch: a per-element
+    // `ptr ? load : skip` select makes hipcc branch around every load and
+    // wait vmcnt(0) per element (serialized L2 round trips).
+    float csv[4] = {1.f, 1.f, 1.f, 1.f};
+    float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+    if (csp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+    }
+    if (biasp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+    }
 #pragma unroll
     for (int i16 = 0; i16 < 4; i16++) {
 #pragma unroll
         for (int r = 0; r < 4; r++) {
             int li = wm + i16 * 16 + kq * 4 + r;           // 0..127
-            int gi = m0 + li;
-            const ushort_t* auxrow =
-                auxp ? auxp + (long)gi * p.aux_ld : nullptr;
+            long gi = m0 + li;
+            float vv[4];
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++)
+                vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+            if (auxp) {
+                const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                ushort_t av[4];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    vv[j16] *= gelu_grad_f(bf2f(av[j16]));
+            }
 #pragma unroll
             for (int j16 = 0; j16 < 4; j16++) {
                 int lj = wn + j16 * 16 + lrow;             // 0..127
-                int gj = n0 + lj;
-                float v = acc[i16][j16][r] * p.alpha;
-                if (csp) v *= csp[gj];
-                if (auxrow) v *= gelu_grad_f(bf2f(auxrow[gj]));
-                if (biasp) v += bf2f(biasp[gj]);
-                scratch[li * EPI_LDS_ROW + lj] = f2bf(v);
+                scratch[li * EPI_LDS_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
             }
         }
     }

@@ -507,23 +507,42 @@ __global__ __launch_bounds__(NT2) void gemm_nt_fast2_kernel(GemmParams p) {
 #pragma unroll
     for (int half = 0; half < 2; half++) {
         if (wm == half * 128) {
+            float csv[4] = {1.f, 1.f, 1.f, 1.f};
+            float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+            if (csp) {
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+            }
+            if (biasp) {
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+            }
 #pragma unroll
             for (int i16 = 0; i16 < 8; i16++) {
 #pragma unroll
                 for (int r = 0; r < 4; r++) {
                     int li = i16 * 16 + kq * 4 + r;        // 0..127
                     long gi = m0 + half * 128 + li;
-                    const ushort_t* auxrow =
-                        auxp ? auxp + gi * p.aux_ld : nullptr;
+                    float vv[4];
+#pragma unroll
+                    for (int j16 = 0; j16 < 4; j16++)
+                        vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+                    if (auxp) {
+                        const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                        ushort_t av[4];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            vv[j16] *= gelu_grad_f(bf2f(av[j16]));
+                    }
 #pragma unroll
                     for (int j16 = 0; j16 < 4; j16++) {
                         int lj = wn + j16 * 16 + lrow;     // 0..255
-                        int gj = n0 + lj;
-                        float v = acc[i16][j16][r] * p.alpha;
-                        if (csp) v *= csp[gj];
-                        if (auxrow) v *= gelu_grad_f(bf2f(auxrow[gj]));
-                        if (biasp) v += bf2f(biasp[gj]);
-                        smem[li * EPI2_ROW + lj] = f2bf(v);
+                        smem[li * EPI2_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
                     }
                 }
             }
