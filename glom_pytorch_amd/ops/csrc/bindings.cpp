@@ -64,10 +64,7 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
             p.ws = ws.data_ptr<float>();
         }
     }
-    // 256^2 tile only pays with an 8-phase schedule (guide tile-space
-    // table); with the 2-barrier loop it regressed, so it stays off until
-    // the phase-pipelined variant lands.
-    const bool nt2 = false;
+    const bool nt2 = nt_fast && p.M % 256 == 0 && p.N % 256 == 0;
     if (nt2)
         launch_gemm_nt_fast2(p, s);
     else if (nt_fast)

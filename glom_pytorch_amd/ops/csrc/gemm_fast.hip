@@ -400,9 +400,12 @@ __device__ __forceinline__ void stage_glds256(
 #define EPI2_ROW 264   // 256 cols + 8 pad
 
 __global__ __launch_bounds__(NT2) void gemm_nt_fast2_kernel(GemmParams p) {
-    __shared__ ushort_t smem[4 * BM2 * FBK];   // 128 KiB: 2 bufs x (A|B)
-    ushort_t* As0 = smem;
-    ushort_t* Bs0 = smem + 2 * BM2 * FBK;
+    // 8 half-tile slots (16 KiB each = 128 rows x 64 k bf16): a 2-deep
+    // double buffer at HALF-tile granularity. Prefetch runs 3-7 halves
+    // ahead with ONE counted s_waitcnt vmcnt(6) per K-tile (2 glds/wave
+    // per half x 3 halves in flight) and raw barriers — the 2-phase
+    // drain-every-tile stall never happens (guide T3+T4).
+    __shared__ ushort_t smem[8 * 128 * FBK];
 
     const int pid = blockIdx.z;
     int nwg = gridDim.x * gridDim.y;
@@ -430,47 +433,92 @@ __global__ __launch_bounds__(NT2) void gemm_nt_fast2_kernel(GemmParams p) {
     f32x4 acc[8][4] = {};
 
     const int nk = p.K / FBK;
-    int cur = 0;
-    stage_glds256(As0, Ap, lda, m0, 0, wid, lane);
-    stage_glds256(Bs0, Bp, ldb, n0, 0, wid, lane);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    const int total_halves = nk * 4;
+
+    // stage one 128-row half-tile (2 glds per wave): h = 0/1 -> A rows
+    // 0-127/128-255, 2/3 -> B rows 0-127/128-255 of this block's tiles
+    auto stage_half = [&](int half_id) {
+        int tile = half_id >> 2;
+        int h = half_id & 3;
+        const ushort_t* src = (h & 2) ? Bp : Ap;
+        long ld = (h & 2) ? ldb : lda;
+        int base0 = ((h & 2) ? n0 : m0) + (h & 1) * 128;
+        ushort_t* lds = smem + (((tile & 1) << 2) | h) * (128 * FBK);
+        int k0 = tile * FBK;
+#pragma unroll
+        for (int c = 0; c < 2; c++) {
+            int chunk = wid * 2 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* gaddr =
+                src + (long)(base0 + row) * ld + k0 + swz8;
+            ushort_t* laddr = lds + chunk * 512;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)gaddr,
+                (__attribute__((address_space(3))) unsigned int*)laddr,
+                16, 0, 0);
+        }
+    };
+
+    // prologue: stage up to 7 halves ahead
+    int next_half = 0;
+    while (next_half < 7 && next_half < total_halves) stage_half(next_half++);
+    if (nk >= 2)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const int a_half = wm >> 7;            // this wave's fixed A half
+    const int b_half = wn >> 7;            // and B half
+    const int arow0 = wm & 127;
+    const int brow0 = wn & 127;
 
     for (int kt = 0; kt < nk; kt++) {
-        if (kt + 1 < nk) {
-            stage_glds256(As0 + (cur ^ 1) * BM2 * FBK, Ap, lda, m0,
-                          (kt + 1) * FBK, wid, lane);
-            stage_glds256(Bs0 + (cur ^ 1) * BM2 * FBK, Bp, ldb, n0,
-                          (kt + 1) * FBK, wid, lane);
-        }
-        const ushort_t* Ab = As0 + cur * BM2 * FBK;
-        const ushort_t* Bb = Bs0 + cur * BM2 * FBK;
+        const ushort_t* Ab = smem + (((kt & 1) << 2) | a_half) * (128 * FBK);
+        const ushort_t* Bb =
+            smem + (((kt & 1) << 2) | (2 | b_half)) * (128 * FBK);
 #pragma unroll
-        for (int s = 0; s < 2; s++) {
-            // B fragments stay live; A fragments stream one row-block at a
-            // time so the 128-reg accumulator + frags fit the 256-VGPR/wave
-            // budget of 2 waves/SIMD (8-wave block)
-            short8 bfr[4];
+        for (int q = 0; q < 4; q++) {      // quadrant: (k-half s, A-row half)
+            int sfr = q >> 1, ih = q & 1;
+            short8 af[4], bfr[4];
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = arow0 + ih * 64 + i * 16 + lrow;
+                int off = (sfr * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[i] = *(const short8*)&Ab[row * FBK + off];
+            }
 #pragma unroll
             for (int j = 0; j < 4; j++) {
-                int row = wn + j * 16 + lrow;
-                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                int row = brow0 + j * 16 + lrow;
+                int off = (sfr * 32 + kq * 8) ^ (swz_row(row) << 3);
                 bfr[j] = *(const short8*)&Bb[row * FBK + off];
             }
+            if (next_half < total_halves) stage_half(next_half++);
+            __builtin_amdgcn_s_barrier();
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+            __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-            for (int i = 0; i < 8; i++) {
-                int row = wm + i * 16 + lrow;
-                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                short8 af = *(const short8*)&Ab[row * FBK + off];
+            for (int i = 0; i < 4; i++)
 #pragma unroll
                 for (int j = 0; j < 4; j++)
-                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af, bfr[j], acc[i][j], 0, 0, 0);
-            }
+                    acc[ih * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bfr[j], acc[ih * 4 + i][j], 0, 0, 0);
+            __builtin_amdgcn_s_setprio(0);
+            __builtin_amdgcn_s_barrier();
         }
-        __syncthreads();
-        cur ^= 1;
+        // tile boundary: next tile's halves must have landed; keep at most
+        // the following tile's 3 halves (6 glds) in flight
+        if (kt + 1 < nk) {
+            if (kt + 2 < nk)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
     }
+    __builtin_amdgcn_s_barrier();
 
     // ---- epilogue: two half-tiles (rows 0-127, 128-255) via LDS ----
     ushort_t* Cp;
