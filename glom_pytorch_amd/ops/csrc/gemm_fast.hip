@@ -627,3 +627,215 @@ void launch_gemm_nt_fast2(const GemmParams& p, hipStream_t stream) {
     dim3 grid(p.N / BN2, p.M / BM2, p.nproblems);
     hipLaunchKernelGGL(gemm_nt_fast2_kernel, grid, dim3(NT2), 0, stream, p);
 }
+
+
+// ---------------------------------------------------------------- //
+// 128x256-tile NT kernel: 8 waves (512 threads), per-wave 64x64, BK=64
+// double-buffered glds, ONE barrier per K-step (the proven 2-phase loop).
+// 2x the per-block work of the 128^2 kernel at 96 KiB LDS — occupancy
+// stays at 2 waves/SIMD while the per-block prologue/epilogue halves
+// relative to the K=512 GLOM shapes that dominate the step.
+
+#define NT3 512
+#define BN3 256
+
+__global__ __launch_bounds__(NT3) void gemm_nt_fast3_kernel(GemmParams p) {
+    __shared__ ushort_t smem[2 * (128 + 256) * FBK];   // 96 KiB
+    const int abuf = 128 * FBK, bbuf = 256 * FBK, stride = abuf + bbuf;
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN3;
+    const int m0 = (bid % gridDim.y) * BM;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 2) * 64;
+    const int wn = (wid & 3) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    // A: 16 chunks over 8 waves (2 each); B: 32 chunks (4 each)
+    auto stage = [&](int buf, int k0) {
+        ushort_t* Al = smem + buf * stride;
+        ushort_t* Bl = Al + abuf;
+#pragma unroll
+        for (int c = 0; c < 2; c++) {
+            int chunk = wid * 2 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g = Ap + (long)(m0 + row) * lda + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (Al + chunk * 512), 16, 0, 0);
+        }
+#pragma unroll
+        for (int c = 0; c < 4; c++) {
+            int chunk = wid * 4 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g = Bp + (long)(n0 + row) * ldb + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (Bl + chunk * 512), 16, 0, 0);
+        }
+    };
+
+    const int nk = p.K / FBK;
+    int cur = 0;
+    stage(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+    for (int kt = 0; kt < nk; kt++) {
+        if (kt + 1 < nk) stage(cur ^ 1, (kt + 1) * FBK);
+        const ushort_t* Al = smem + cur * stride;
+        const ushort_t* Bl = Al + abuf;
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&Al[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bl[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+        cur ^= 1;
+    }
+
+    // ---- epilogue: 128x256 tile via LDS, full-line stores ----
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    ushort_t* out2p = nullptr;
+    if (p.epilogue == EPI_GELU_PAIR)
+        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout;
+
+    float csv[4] = {1.f, 1.f, 1.f, 1.f};
+    float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+    if (csp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+    }
+    if (biasp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+    }
+#pragma unroll
+    for (int i16 = 0; i16 < 4; i16++) {
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            int li = wm + i16 * 16 + kq * 4 + r;           // 0..127
+            long gi = m0 + li;
+            float vv[4];
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++)
+                vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+            if (auxp) {
+                const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                ushort_t av[4];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    vv[j16] *= gelu_grad_f(bf2f(av[j16]));
+            }
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int lj = wn + j16 * 16 + lrow;             // 0..255
+                smem[li * EPI2_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
+            }
+        }
+    }
+    __syncthreads();
+    {
+        int t = threadIdx.x;              // 512 threads: 128 rows x 4 qtrs
+        int li = t >> 2;
+        int qt = (t & 3) * 64;
+        long gi = m0 + li;
+        ushort_t* crow = Cp + gi * ldc + n0 + qt;
+        ushort_t* orow = out2p ? out2p + gi * p.out2_ld + n0 + qt : nullptr;
+        const ushort_t* srow = smem + li * EPI2_ROW + qt;
+#pragma unroll
+        for (int c = 0; c < 8; c++)
+            *(uint4v*)(crow + c * 8) = *(const uint4v*)(srow + c * 8);
+        if (orow) {
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x, g;
+                x.v = *(const uint4v*)(srow + c * 8);
+                if (p.epilogue == 10) {
+                    g.v = x.v;
+                } else {
+#pragma unroll
+                    for (int e = 0; e < 8; e++)
+                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+                }
+                *(uint4v*)(orow + c * 8) = g.v;
+            }
+        }
+    }
+}
+
+void launch_gemm_nt_fast3(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN3, p.M / BM, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast3_kernel, grid, dim3(NT3), 0, stream, p);
+}
