@@ -483,7 +483,7 @@ double bench_gemm(int64_t M, int64_t N, int64_t K, int64_t layout,
         if (epilogue == EPI_GELUGRAD) {
             p.aux_base = aux.data_ptr(); p.aux_sin = M * N; p.aux_ld = N;
         }
-        if (epilogue == EPI_GELU_PAIR || epilogue == 10) {
+        if (epilogue == EPI_GELU_PAIR || epilogue == 10 || epilogue == 12) {
             p.bias_base = bias.data_ptr(); p.bias_sin = N; p.has_bias = 1;
             p.out2 = out2.data_ptr(); p.out2_sin = M * N; p.out2_ld = N;
         }

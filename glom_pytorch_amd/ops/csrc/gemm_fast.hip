@@ -824,6 +824,12 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast3_kernel(GemmParams p) {
                 x.v = *(const uint4v*)(srow + c * 8);
                 if (p.epilogue == 10) {
                     g.v = x.v;
+                } else if (p.epilogue == 12) {      // debug: poly, no trans
+#pragma unroll
+                    for (int e = 0; e < 8; e++) {
+                        float xv = bf2f(x.u[e]);
+                        g.u[e] = f2bf(xv * (0.5f + 0.1f * xv));
+                    }
                 } else {
 #pragma unroll
                     for (int e = 0; e < 8; e++)

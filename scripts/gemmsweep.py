@@ -21,3 +21,5 @@ for name, M,N,K,P,lay,epi in cases:
     print(f"{name:26s} {ms*1e3:8.1f} us  {tf(M,N,K,P,ms):6.1f} TF")
 ms = ext.bench_gemm(16384,2048,512,NT,6,10,20)
 print(f"{'up pair NOGELU dbg':26s} {ms*1e3:8.1f} us  {tf(16384,2048,512,6,ms):6.1f} TF")
+ms = ext.bench_gemm(16384,2048,512,NT,6,12,20)
+print(f"{'up pair POLY dbg':26s} {ms*1e3:8.1f} us  {tf(16384,2048,512,6,ms):6.1f} TF")
