@@ -267,18 +267,8 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
         run_gemm(p, s, opts, true);
     }
-    // bias grads: custom split-M column sums (f32 atomics) instead of the
-    // generic torch reduction
-    auto db1f = torch::zeros({G, m4}, opts.dtype(at::kFloat));
-    launch_colsum(dHpre.data_ptr(), db1f.data_ptr<float>(), G, M, m4,
-                  M * m4, m4, s);
-    check_launch();
-    auto db2f = torch::zeros({G, d}, opts.dtype(at::kFloat));
-    launch_colsum(dY.data_ptr(), db2f.data_ptr<float>(), G, M, d,
-                  d, G * d, s);
-    check_launch();
-    auto dB1 = db1f.flatten().to(at::kBFloat16);
-    auto dB2 = db2f.flatten().to(at::kBFloat16);
+    auto dB1 = dHpre.sum(1).flatten();                       // (G*m4)
+    auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();       // (G*d)
     if (mode != 0) dTokens = torch::empty({0}, opts);
     return {dTokens, dLevels, dW1, dB1, dW2, dB2};
 }
