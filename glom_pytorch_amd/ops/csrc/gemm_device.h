@@ -155,9 +155,12 @@ __device__ __forceinline__ void gemm_epilogue_lds(
 #pragma unroll
                 for (int j16 = 0; j16 < 4; j16++)
                     av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+                float gx[4], gy[4];
 #pragma unroll
-                for (int j16 = 0; j16 < 4; j16++)
-                    vv[j16] *= gelu_grad_f(bf2f(av[j16]));
+                for (int j16 = 0; j16 < 4; j16++) gx[j16] = bf2f(av[j16]);
+                gelu_grad_vec<4>(gx, gy);
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++) vv[j16] *= gy[j16];
             }
 #pragma unroll
             for (int j16 = 0; j16 < 4; j16++) {
@@ -188,9 +191,12 @@ __device__ __forceinline__ void gemm_epilogue_lds(
                 if (p.epilogue == 10) {       // debug: pair without gelu
                     g.v = x.v;
                 } else {
+                    float xin[8], yv[8];
 #pragma unroll
-                    for (int e = 0; e < 8; e++)
-                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+                    for (int e = 0; e < 8; e++) xin[e] = bf2f(x.u[e]);
+                    gelu_f_vec<8>(xin, yv);
+#pragma unroll
+                    for (int e = 0; e < 8; e++) g.u[e] = f2bf(yv[e]);
                 }
                 *(uint4v*)(orow + c * 8) = g.v;
             }
