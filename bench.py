@@ -35,6 +35,9 @@ def parse_args():
                    default="native",
                    help="native = CDNA4 HIP engine; eager/compile = stock "
                         "PyTorch-ROCm running the same math (comparison)")
+    p.add_argument("--mode", choices=["train", "infer"], default="train",
+                   help="train = denoising fwd+bwd+AdamW (headline); "
+                        "infer = no-grad forward under hipGraph replay")
     return p.parse_args()
 
 
@@ -61,8 +64,6 @@ def main():
     if args.impl == "compile":
         model = torch.compile(model)
 
-    trainer = DenoisingTrainer(model, distributed=distributed)
-
     img = torch.randn(args.batch, 3, args.image_size, args.image_size,
                       device=dev, dtype=torch.bfloat16)
 
@@ -71,13 +72,28 @@ def main():
             dist.barrier()
         torch.cuda.synchronize()
 
+    if args.mode == "infer":
+        if args.impl == "native":
+            model.enable_graphs()
+        loss = 0.0
+
+        def step(x, iters):
+            with torch.no_grad():
+                model(x, iters=iters)
+            return 0.0
+    else:
+        trainer = DenoisingTrainer(model, distributed=distributed)
+
+        def step(x, iters):
+            return trainer.step(x, iters=iters)
+
     for _ in range(args.warmup):
-        trainer.step(img, iters=args.iters)
+        step(img, args.iters)
 
     sync_all()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        loss = trainer.step(img, iters=args.iters)
+        loss = step(img, args.iters)
     sync_all()
     elapsed = time.perf_counter() - t0
 
@@ -110,7 +126,9 @@ def main():
                 "image_size": args.image_size,
                 "patch_size": args.patch_size,
                 "iters": args.iters,
-                "objective": "denoising-mse fwd+bwd+adamw",
+                "objective": ("denoising-mse fwd+bwd+adamw"
+                              if args.mode == "train"
+                              else "inference forward (hipGraph replay)"),
                 "impl": args.impl,
                 "parallelism": f"dp{n_gpus}",
             },
