@@ -300,7 +300,8 @@ std::vector<torch::Tensor> consensus_fwd(
     check_launch();
 
     auto probs = torch::empty({B, L, N, N}, opts);
-    // scores[i,j] = (q_i . k_j) * rnorm_j * d^-0.5
+    const bool fused_sm = (N == 256) && (d % 64 == 0) && lds_ok;
+    // scores[i,j] = (q_i . k_j) * rnorm_j * d^-0.5, then masked row softmax
     {
         GemmParams p = base_params(N, N, d, LAYOUT_NT, P, L,
                                    (float)std::pow((double)d, -0.5));
@@ -311,12 +312,21 @@ std::vector<torch::Tensor> consensus_fwd(
         p.Cld = N;
         p.colscale_base = rnorm.data_ptr<float>();
         p.cs_sin = N; p.cs_sout = L * N; p.has_colscale = 1;
-        run_gemm(p, s, opts, lds_ok);
+        if (fused_sm) {
+            // the nt3 tile spans whole rows: softmax fuses into the epilogue
+            p.epilogue = EPI_SOFTMAX;
+            p.self_mask = attend_self ? 0 : 1;
+            p.nlmask = mask;
+            p.splitk = 1;
+            launch_gemm_nt_fast3(p, s);
+            check_launch();
+        } else {
+            run_gemm(p, s, opts, lds_ok);
+            launch_softmax_fwd(probs.data_ptr(), probs.data_ptr(), mask,
+                               (int)P, (int)N, attend_self ? 0 : 1, s);
+            check_launch();
+        }
     }
-    // masked softmax in place
-    launch_softmax_fwd(probs.data_ptr(), probs.data_ptr(), mask, (int)P,
-                       (int)N, attend_self ? 0 : 1, s);
-    check_launch();
     // out[i,:] = sum_j P[i,j] * levels[j,:]
     auto out = torch::empty({B, N, L, d}, opts);
     {
@@ -346,9 +356,34 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
     const bool* mask = nullptr;
     if (mask_opt.has_value()) mask = mask_opt.value().data_ptr<bool>();
 
-    // dP[i,j] = dOut_i . v_j
-    auto dP = torch::empty({B, L, N, N}, opts);
-    {
+    // dP[i,j] = dOut_i . v_j, then row softmax backward -> dS, dSr
+    const bool fused_sm = (N == 256) && (d % 64 == 0) && lds_ok;
+    auto dS = torch::empty({B, L, N, N}, opts);
+    auto dSr = torch::empty({B, L, N, N}, opts);
+    torch::Tensor dP;
+    if (fused_sm) {
+        GemmParams p = base_params(N, N, d, LAYOUT_NT, P, L, 1.0f);
+        p.A.base = dOut.data_ptr(); p.A.sin = d; p.A.sout = N * L * d;
+        p.A.ld = L * d;
+        p.B.base = levels.data_ptr(); p.B.sin = d; p.B.sout = N * L * d;
+        p.B.ld = L * d;
+        p.Cbase = dS.data_ptr(); p.Csin = N * N; p.Csout = L * N * N;
+        p.Cld = N;
+        p.epilogue = EPI_SMBWD;
+        p.aux_base = probs.data_ptr();
+        p.aux_sin = N * N; p.aux_sout = L * N * N; p.aux_ld = N;
+        p.out2 = dSr.data_ptr();
+        p.out2_sin = N * N; p.out2_sout = L * N * N; p.out2_ld = N;
+        p.colscale_base = rnorm.data_ptr<float>();
+        p.cs_sin = N; p.cs_sout = L * N; p.has_colscale = 1;
+        p.self_mask = attend_self ? 0 : 1;
+        p.nlmask = mask;
+        p.alpha2 = (float)std::pow((double)d, -0.5);
+        p.splitk = 1;
+        launch_gemm_nt_fast3(p, s);
+        check_launch();
+    } else {
+        dP = torch::empty({B, L, N, N}, opts);
         GemmParams p = base_params(N, N, d, LAYOUT_NT, P, L, 1.0f);
         p.A.base = dOut.data_ptr(); p.A.sin = d; p.A.sout = N * L * d;
         p.A.ld = L * d;
@@ -357,14 +392,13 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
         p.Cbase = dP.data_ptr(); p.Csin = N * N; p.Csout = L * N * N;
         p.Cld = N;
         run_gemm(p, s, opts, lds_ok);
+        launch_softmax_bwd(probs.data_ptr(), dP.data_ptr(),
+                           rnorm.data_ptr<float>(), dS.data_ptr(),
+                           dSr.data_ptr(), mask, (int)P, (int)N,
+                           attend_self ? 0 : 1,
+                           (float)std::pow((double)d, -0.5), s);
+        check_launch();
     }
-    auto dS = torch::empty_like(dP);
-    auto dSr = torch::empty_like(dP);
-    launch_softmax_bwd(probs.data_ptr(), dP.data_ptr(),
-                       rnorm.data_ptr<float>(), dS.data_ptr(), dSr.data_ptr(),
-                       mask, (int)P, (int)N, attend_self ? 0 : 1,
-                       (float)std::pow((double)d, -0.5), s);
-    check_launch();
 
     // dv[j,:] = sum_i P[i,j] dOut[i,:]
     auto dv = torch::empty({B, N, L, d}, opts);

@@ -5,7 +5,14 @@
 
 enum GemmLayout { LAYOUT_NT = 0, LAYOUT_NN = 1, LAYOUT_TN = 2 };
 enum OpFlags { OP_GELU = 1, OP_POS = 2, OP_TABLE = 4 };
-enum GemmEpilogue { EPI_NONE = 0, EPI_GELUGRAD = 1, EPI_GELU_PAIR = 2 };
+enum GemmEpilogue {
+    EPI_NONE = 0,
+    EPI_GELUGRAD = 1,
+    EPI_GELU_PAIR = 2,
+    // full-row fusions (tile spans the whole row, nt_fast3 only):
+    EPI_SOFTMAX = 3,   // consensus scores -> masked row softmax -> P
+    EPI_SMBWD = 4,     // dP -> row softmax backward -> dS (C), dSr (out2)
+};
 
 #define GEMM_MAX_TABLE 16
 
@@ -36,6 +43,11 @@ struct GemmParams {
     const void* aux_base; long aux_sin, aux_sout, aux_ld;
 
     const void* pos; long pos_ld;
+
+    // attention-fusion extras (EPI_SOFTMAX / EPI_SMBWD)
+    const void* nlmask;     // optional (N,N) bool non-local mask
+    int self_mask;          // fill diagonal with -5e-4 before softmax
+    float alpha2;           // d^-0.5 applied inside EPI_SMBWD
 
     // split-K (reduction split): when > 1, the kernel writes per-slice f32
     // partials into ws[slice][problem][M][N] and gemm_finish sums them into

@@ -768,7 +768,7 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast3_kernel(GemmParams p) {
 
     float csv[4] = {1.f, 1.f, 1.f, 1.f};
     float bvv[4] = {0.f, 0.f, 0.f, 0.f};
-    if (csp) {
+    if (csp && p.epilogue != EPI_SMBWD) {
 #pragma unroll
         for (int j16 = 0; j16 < 4; j16++)
             csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
@@ -809,6 +809,119 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast3_kernel(GemmParams p) {
         }
     }
     __syncthreads();
+
+    if (p.epilogue == EPI_SOFTMAX || p.epilogue == EPI_SMBWD) {
+        // Full-row fusions: this tile holds complete rows (N == 256), so
+        // the masked row softmax (fwd) / softmax backward (bwd) runs here
+        // instead of a separate kernel + a global-memory round trip.
+        // 4 consecutive threads (one quartet, same wave) own one row.
+        int t = threadIdx.x;
+        int li = t >> 2;
+        int qt = (t & 3) * 64;
+        long gi = m0 + li;
+        const bool* mrow =
+            p.nlmask ? (const bool*)p.nlmask + gi * p.N + n0 + qt : nullptr;
+        ushort_t* srow = smem + li * EPI2_ROW + qt;
+        const float SELF = bf2f(f2bf(-5e-4f));
+        const float NEG = -3.3895314e38f;
+
+        if (p.epilogue == EPI_SOFTMAX) {
+            // pass 1: masks + quartet max
+            float mx = -INFINITY;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++) {
+                    int gj = n0 + qt + c * 8 + e;
+                    float v = bf2f(x.u[e]);
+                    if (p.self_mask && gj == gi) v = SELF;
+                    if (mrow && mrow[c * 8 + e]) v = NEG;
+                    x.u[e] = f2bf(v);
+                    mx = fmaxf(mx, v);
+                }
+                *(uint4v*)(srow + c * 8) = x.v;   // masked values back
+            }
+            mx = fmaxf(mx, __shfl_xor(mx, 1));
+            mx = fmaxf(mx, __shfl_xor(mx, 2));
+            // pass 2: exp + quartet sum (store unnormalized bf16)
+            float sum = 0.f;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++) {
+                    float ex = __expf(bf2f(x.u[e]) - mx);
+                    sum += ex;
+                    x.u[e] = f2bf(ex);
+                }
+                *(uint4v*)(srow + c * 8) = x.v;
+            }
+            sum += __shfl_xor(sum, 1);
+            sum += __shfl_xor(sum, 2);
+            float inv = 1.0f / sum;
+            // pass 3: normalize + store P
+            ushort_t* prow = Cp + gi * ldc + n0 + qt;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++)
+                    x.u[e] = f2bf(bf2f(x.u[e]) * inv);
+                *(uint4v*)(prow + c * 8) = x.v;
+            }
+        } else {
+            // EPI_SMBWD: dS = alpha2 * P * (dP - sum(P*dP)), masked -> 0;
+            // dSr = dS * rnorm[j]. P comes via aux, rnorm via colscale
+            // pointers (phase 1 ran with alpha=1, no colscale).
+            const ushort_t* Prow = (const ushort_t*)p.aux_base
+                + (long)(pid % p.nInner) * p.aux_sin
+                + (long)(pid / p.nInner) * p.aux_sout + gi * p.aux_ld
+                + n0 + qt;
+            const float* rn = (const float*)p.colscale_base
+                + (long)(pid % p.nInner) * p.cs_sin
+                + (long)(pid / p.nInner) * p.cs_sout + n0 + qt;
+            float tsum = 0.f;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } dp, pp;
+                dp.v = *(const uint4v*)(srow + c * 8);
+                pp.v = *(const uint4v*)(Prow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++)
+                    tsum += bf2f(pp.u[e]) * bf2f(dp.u[e]);
+            }
+            tsum += __shfl_xor(tsum, 1);
+            tsum += __shfl_xor(tsum, 2);
+            ushort_t* dsrow = Cp + gi * ldc + n0 + qt;
+            ushort_t* dsr_row = (ushort_t*)p.out2
+                + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout + gi * p.out2_ld
+                + n0 + qt;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } dp, pp, o1, o2;
+                dp.v = *(const uint4v*)(srow + c * 8);
+                pp.v = *(const uint4v*)(Prow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++) {
+                    int gj = n0 + qt + c * 8 + e;
+                    float v = p.alpha2 * bf2f(pp.u[e])
+                              * (bf2f(dp.u[e]) - tsum);
+                    if (p.self_mask && gj == gi) v = 0.f;
+                    if (mrow && mrow[c * 8 + e]) v = 0.f;
+                    o1.u[e] = f2bf(v);
+                    o2.u[e] = f2bf(v * rn[c * 8 + e]);
+                }
+                *(uint4v*)(dsrow + c * 8) = o1.v;
+                *(uint4v*)(dsr_row + c * 8) = o2.v;
+            }
+        }
+        return;
+    }
     {
         int t = threadIdx.x;              // 512 threads: 128 rows x 4 qtrs
         int li = t >> 2;

@@ -221,3 +221,40 @@ def test_parity_fast_tile_shapes():
         cos = torch.nn.functional.cosine_similarity(
             g32.flatten(), gbf.flatten(), dim=0).item()
         assert cos > 0.99, (n32, cos)
+
+
+@pytest.mark.parametrize("attend_self,radius", [(False, 0), (True, 0),
+                                                (False, 3)])
+def test_consensus_fused_softmax_n256(attend_self, radius):
+    """N=256 rides the fused scores+softmax / dP+softmax-bwd epilogues;
+    verify against fp32 einsum math including the masked variants."""
+    import math
+    import torch.nn.functional as F
+    from glom_pytorch_amd.ops.functional import ConsensusFn
+    torch.manual_seed(5)
+    B, N, L, d = 1, 256, 2, 64
+    lv = torch.randn(B, N, L, d, device=DEV, dtype=torch.bfloat16,
+                     requires_grad=True)
+    mask = None
+    if radius:
+        side = 16
+        hh, ww = torch.meshgrid(torch.arange(side), torch.arange(side),
+                                indexing="ij")
+        coords = torch.stack((hh, ww)).float().view(2, -1).t()
+        mask = (torch.cdist(coords, coords) > radius).view(1, N, N).to(DEV)
+    out = ConsensusFn.apply(lv, attend_self, mask)
+    g = torch.randn_like(out)
+    (dlev,) = torch.autograd.grad(out, lv, g)
+
+    x = lv.detach().float().requires_grad_(True)
+    sim = torch.einsum("bild,bjld->blij", x,
+                       F.normalize(x, dim=-1)) / math.sqrt(d)
+    if not attend_self:
+        eye = torch.eye(N, device=DEV, dtype=torch.bool)
+        sim = sim.masked_fill(eye.view(1, 1, N, N), -5e-4)
+    if mask is not None:
+        sim = sim.masked_fill(mask.view(1, 1, N, N), -3.3895e38)
+    ref = torch.einsum("blij,bjld->bild", sim.softmax(-1), x)
+    (dref,) = torch.autograd.grad(ref, x, g.float())
+    assert _rel_err(out, ref) < 1e-2
+    assert _rel_err(dlev, dref) < 3e-2
