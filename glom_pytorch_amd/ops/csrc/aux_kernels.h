@@ -21,3 +21,5 @@ void launch_colsum(const void* in, float* out, long P, long M, long C,
                    long sin, long ld, hipStream_t s);
 void launch_mix_bwd(const void* dout, void* dmix, void* dtd, long total,
                     int L, int d, hipStream_t s);
+void launch_add4(const void* a, const void* b, const void* c, const void* d,
+                 void* out, long total, hipStream_t s);

@@ -423,3 +423,31 @@ void launch_colsum(const void* in, float* out, long P, long M, long C,
     hipLaunchKernelGGL(k_colsum, grid, dim3(NTHREADS), 0, s,
                        (const ushort_t*)in, out, M, C, sin, ld, mchunks);
 }
+
+// out = a + b + c + d, elementwise bf16 (the four levels-gradient
+// contributions of one GLOM iteration summed in one pass)
+__global__ __launch_bounds__(NTHREADS) void k_add4(
+        const ushort_t* __restrict__ a, const ushort_t* __restrict__ b,
+        const ushort_t* __restrict__ c, const ushort_t* __restrict__ d,
+        ushort_t* __restrict__ out, long total) {
+    long i8 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
+    if (i8 >= total) return;
+    union { uint4v v; ushort_t u[8]; } va, vb, vc, vd, o;
+    va.v = *(const uint4v*)(a + i8);
+    vb.v = *(const uint4v*)(b + i8);
+    vc.v = *(const uint4v*)(c + i8);
+    vd.v = *(const uint4v*)(d + i8);
+#pragma unroll
+    for (int e = 0; e < 8; e++)
+        o.u[e] = f2bf(bf2f(va.u[e]) + bf2f(vb.u[e]) + bf2f(vc.u[e])
+                      + bf2f(vd.u[e]));
+    *(uint4v*)(out + i8) = o.v;
+}
+
+void launch_add4(const void* a, const void* b, const void* c, const void* d,
+                 void* out, long total, hipStream_t s) {
+    hipLaunchKernelGGL(k_add4, dim3(cdiv(total / 8, NTHREADS)),
+                       dim3(NTHREADS), 0, s, (const ushort_t*)a,
+                       (const ushort_t*)b, (const ushort_t*)c,
+                       (const ushort_t*)d, (ushort_t*)out, total);
+}

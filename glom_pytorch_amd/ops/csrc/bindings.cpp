@@ -206,7 +206,12 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         check_launch();
     }
 
-    // dHpre = (dY_g @ W2_g) * gelu'(Hpre_g)  -- NT with W2^T
+    // dHpre = (dY_g @ W2_g) * gelu'(Hpre_g)  -- NT with W2^T; when the
+    // 128x256 kernel serves it, the bias grad dB1 = colsum(dHpre) fuses
+    // into its epilogue (saves a full re-read of the 400MB dHpre)
+    const bool dh_nt3 = (M % 128 == 0) && (m4 % 256 == 0) && (m4 >= 1024)
+                        && (d % 64 == 0);
+    torch::Tensor db1f;
     {
         GemmParams p = base_params(M, m4, d, LAYOUT_NT, G, G, 1.0f);
         p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
@@ -214,6 +219,11 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         p.Cbase = dHpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
         p.epilogue = EPI_GELUGRAD;
         p.aux_base = Hpre.data_ptr(); p.aux_sin = M * m4; p.aux_ld = m4;
+        if (dh_nt3) {
+            db1f = torch::zeros({G, m4}, opts.dtype(at::kFloat));
+            p.colsum_out = db1f.data_ptr<float>();
+            p.colsum_sin = m4;
+        }
         run_gemm(p, s, opts, true);
     }
     // dX_g = dHpre_g @ W1_g -- NT with W1^T, scattered into level slices
@@ -267,7 +277,8 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
         run_gemm(p, s, opts, true);
     }
-    auto dB1 = dHpre.sum(1).flatten();                       // (G*m4)
+    auto dB1 = dh_nt3 ? db1f.flatten().to(at::kBFloat16)
+                      : dHpre.sum(1).flatten();              // (G*m4)
     auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();       // (G*d)
     if (mode != 0) dTokens = torch::empty({0}, opts);
     return {dTokens, dLevels, dW1, dB1, dW2, dB2};
@@ -474,6 +485,55 @@ std::vector<torch::Tensor> level_mix_bwd(torch::Tensor dout) {
     return {dmix, dtd};
 }
 
+
+// ------------------------------------------------------------------ //
+// One full GLOM iteration as a single op pair (reference
+// glom_pytorch.py:131-145): forward chains bottom-up, top-down, consensus
+// and the level mix in one extension call; backward additionally sums the
+// four levels-gradient contributions in one fused pass instead of three
+// autograd accumulations.
+
+std::vector<torch::Tensor> glom_step_fwd(
+        torch::Tensor tokens, torch::Tensor levels, torch::Tensor pos,
+        torch::Tensor bw1, torch::Tensor bb1, torch::Tensor bw2,
+        torch::Tensor bb2, torch::Tensor tw1, torch::Tensor tb1,
+        torch::Tensor tw2, torch::Tensor tb2, bool attend_self,
+        c10::optional<torch::Tensor> mask) {
+    auto bu = grouped_ff_fwd(tokens, levels, c10::nullopt, bw1, bb1, bw2,
+                             bb2, 0);
+    auto td = grouped_ff_fwd(c10::nullopt, levels, pos, tw1, tb1, tw2,
+                             tb2, 1);
+    auto at = consensus_fwd(levels, attend_self, mask);
+    auto out = level_mix_fwd(levels, bu[0], td[0], at[0]);
+    return {out, bu[1], bu[2], td[1], td[2], at[1], at[2]};
+}
+
+std::vector<torch::Tensor> glom_step_bwd(
+        torch::Tensor dnew, torch::Tensor tokens, torch::Tensor levels,
+        torch::Tensor pos, torch::Tensor bw1, torch::Tensor bw2,
+        torch::Tensor tw1, torch::Tensor tw2, torch::Tensor buHpre,
+        torch::Tensor buHact, torch::Tensor tdHpre, torch::Tensor tdHact,
+        torch::Tensor probs, torch::Tensor rnorm, bool attend_self,
+        c10::optional<torch::Tensor> mask) {
+    const int64_t L = levels.size(2);
+    auto mix = level_mix_bwd(dnew);        // {dmix, dtd}
+    auto bu = grouped_ff_bwd(mix[0], tokens, levels, c10::nullopt, bw1,
+                             bw2, buHpre, buHact, 0);
+    auto td = grouped_ff_bwd(mix[1], c10::nullopt, levels, pos, tw1, tw2,
+                             tdHpre, tdHact, 1);
+    auto dAttn = consensus_bwd(mix[0], levels, probs, rnorm, attend_self,
+                               mask);
+    auto dLevels = torch::empty_like(levels);
+    launch_add4(mix[0].data_ptr(), bu[1].data_ptr(), td[1].data_ptr(),
+                dAttn.data_ptr(), dLevels.data_ptr(), levels.numel(),
+                cur_stream());
+    check_launch();
+    auto dPos = td[1].narrow(2, 1, L - 1)
+                    .sum(std::vector<int64_t>{0, 2});   // (N, d)
+    return {bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
+            td[2], td[3], td[4], td[5]};
+}
+
 // Synthetic single-GEMM microbenchmark for kernel tuning (used by
 // scripts/gemmbench.py; not part of the model path).
 double bench_gemm(int64_t M, int64_t N, int64_t K, int64_t layout,
@@ -539,6 +599,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("consensus_bwd", &consensus_bwd, "consensus attention backward");
     m.def("level_mix_fwd", &level_mix_fwd, "level mix forward");
     m.def("level_mix_bwd", &level_mix_bwd, "level mix backward");
+    m.def("glom_step_fwd", &glom_step_fwd, "full GLOM iteration forward");
+    m.def("glom_step_bwd", &glom_step_bwd, "full GLOM iteration backward");
     m.def("build_info", &build_info);
     m.def("bench_gemm", &bench_gemm, "raw GEMM microbench (tuning only)");
 }

@@ -44,6 +44,11 @@ struct GemmParams {
 
     const void* pos; long pos_ld;
 
+    // optional fused column sum of C over the row axis (bias gradient):
+    // nt_fast3 adds its tile's column sums into colsum_out[p][col] (f32,
+    // caller-zeroed) during the epilogue, saving a full re-read of C.
+    float* colsum_out; long colsum_sin;
+
     // attention-fusion extras (EPI_SOFTMAX / EPI_SMBWD)
     const void* nlmask;     // optional (N,N) bool non-local mask
     int self_mask;          // fill diagonal with -5e-4 before softmax

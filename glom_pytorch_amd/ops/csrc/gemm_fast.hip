@@ -958,6 +958,18 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast3_kernel(GemmParams p) {
             }
         }
     }
+    if (p.colsum_out) {
+        // fused bias-grad: this tile's column sums (C values still live in
+        // the LDS staging image) accumulated into the f32 output
+        float* outp = p.colsum_out + (long)(pid % p.nInner) * p.colsum_sin;
+        int c = threadIdx.x;
+        if (c < BN3) {
+            float ssum = 0.f;
+            for (int r = 0; r < 128; r++)
+                ssum += bf2f(smem[r * EPI2_ROW + c]);
+            atomicAdd(&outp[n0 + c], ssum);
+        }
+    }
 }
 
 void launch_gemm_nt_fast3(const GemmParams& p, hipStream_t stream) {

@@ -86,15 +86,46 @@ class LevelMixFn(torch.autograd.Function):
         return dmix, dmix, dtd, dmix
 
 
+class GlomStepFn(torch.autograd.Function):
+    """One full GLOM iteration (reference glom_pytorch.py:131-145) as a
+    single autograd node: forward chains bottom-up / top-down / consensus /
+    level-mix in one extension call; backward sums the four levels-gradient
+    contributions in one fused kernel instead of three engine
+    accumulations."""
+
+    @staticmethod
+    def forward(ctx, tokens, levels, pos, bw1, bb1, bw2, bb2,
+                tw1, tb1, tw2, tb2, attend_self, mask):
+        ext = _load_extension()
+        out, bhp, bha, thp, tha, probs, rnorm = ext.glom_step_fwd(
+            tokens, levels, pos, bw1, bb1, bw2, bb2, tw1, tb1, tw2, tb2,
+            attend_self, mask)
+        ctx.save_for_backward(tokens, levels, pos, bw1, bw2, tw1, tw2,
+                              bhp, bha, thp, tha, probs, rnorm, mask)
+        ctx.attend_self = attend_self
+        return out
+
+    @staticmethod
+    def backward(ctx, dnew):
+        ext = _load_extension()
+        (tokens, levels, pos, bw1, bw2, tw1, tw2, bhp, bha, thp, tha,
+         probs, rnorm, mask) = ctx.saved_tensors
+        (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
+         dtw1, dtb1, dtw2, dtb2) = ext.glom_step_bwd(
+            dnew.contiguous(), tokens, levels, pos, bw1, bw2, tw1, tw2,
+            bhp, bha, thp, tha, probs, rnorm, ctx.attend_self, mask)
+        return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
+                dtw1, dtb1, dtw2, dtb2, None, None)
+
+
 def glom_step(model, tokens, levels, pos, mask):
-    w = model.bottom_up.net
-    bu = GroupedFFFn.apply(tokens, levels, None, w[1].weight[..., 0],
-                           w[1].bias, w[3].weight[..., 0], w[3].bias, 0)
-    w = model.top_down.net
-    td = GroupedFFFn.apply(None, levels, pos, w[1].weight[..., 0],
-                           w[1].bias, w[3].weight[..., 0], w[3].bias, 1)
-    cons = ConsensusFn.apply(levels, model.attention.attend_self, mask)
-    return LevelMixFn.apply(levels, bu, td, cons)
+    bw = model.bottom_up.net
+    tw = model.top_down.net
+    return GlomStepFn.apply(
+        tokens, levels, pos,
+        bw[1].weight[..., 0], bw[1].bias, bw[3].weight[..., 0], bw[3].bias,
+        tw[1].weight[..., 0], tw[1].bias, tw[3].weight[..., 0], tw[3].bias,
+        model.attention.attend_self, mask)
 
 
 def glom_forward(model, img, iters, levels=None, return_all=False):
