@@ -177,7 +177,9 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         torch::Tensor dY, c10::optional<torch::Tensor> tokens_opt,
         torch::Tensor levels, c10::optional<torch::Tensor> pos_opt,
         torch::Tensor w1, torch::Tensor w2, torch::Tensor Hpre,
-        torch::Tensor Hact, int64_t mode) {
+        torch::Tensor Hact, int64_t mode,
+        c10::optional<torch::Tensor> w1t_opt = c10::nullopt,
+        c10::optional<torch::Tensor> w2t_opt = c10::nullopt) {
     CHECK_IN(dY); CHECK_IN(levels); CHECK_IN(w1); CHECK_IN(w2); CHECK_IN(Hpre);
     CHECK_IN(Hact);
     const int64_t B = levels.size(0), N = levels.size(1),
@@ -193,8 +195,13 @@ std::vector<torch::Tensor> grouped_ff_bwd(
     torch::Tensor dTokens;
 
     // per-group weight transposes turn the NN data grads into NT GEMMs
-    auto w1t = w1.view({G, m4, d}).transpose(1, 2).contiguous();  // (G,d,m4)
-    auto w2t = w2.view({G, d, m4}).transpose(1, 2).contiguous();  // (G,m4,d)
+    // (callers that run many iterations pass them in, computed once)
+    auto w1t = w1t_opt.has_value()
+        ? w1t_opt.value()
+        : w1.view({G, m4, d}).transpose(1, 2).contiguous();   // (G,d,m4)
+    auto w2t = w2t_opt.has_value()
+        ? w2t_opt.value()
+        : w2.view({G, d, m4}).transpose(1, 2).contiguous();   // (G,m4,d)
 
     torch::Tensor td_in;
     if (mode == 1) {
@@ -514,13 +521,16 @@ std::vector<torch::Tensor> glom_step_bwd(
         torch::Tensor tw1, torch::Tensor tw2, torch::Tensor buHpre,
         torch::Tensor buHact, torch::Tensor tdHpre, torch::Tensor tdHact,
         torch::Tensor probs, torch::Tensor rnorm, bool attend_self,
-        c10::optional<torch::Tensor> mask) {
+        c10::optional<torch::Tensor> mask,
+        c10::optional<torch::Tensor> bw1t, c10::optional<torch::Tensor> bw2t,
+        c10::optional<torch::Tensor> tw1t,
+        c10::optional<torch::Tensor> tw2t) {
     const int64_t L = levels.size(2);
     auto mix = level_mix_bwd(dnew);        // {dmix, dtd}
     auto bu = grouped_ff_bwd(mix[0], tokens, levels, c10::nullopt, bw1,
-                             bw2, buHpre, buHact, 0);
+                             bw2, buHpre, buHact, 0, bw1t, bw2t);
     auto td = grouped_ff_bwd(mix[1], c10::nullopt, levels, pos, tw1, tw2,
-                             tdHpre, tdHact, 1);
+                             tdHpre, tdHact, 1, tw1t, tw2t);
     auto dAttn = consensus_bwd(mix[0], levels, probs, rnorm, attend_self,
                                mask);
     auto dLevels = torch::empty_like(levels);
@@ -594,7 +604,12 @@ std::string build_info() {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("grouped_ff_fwd", &grouped_ff_fwd, "grouped FF forward");
-    m.def("grouped_ff_bwd", &grouped_ff_bwd, "grouped FF backward");
+    m.def("grouped_ff_bwd", &grouped_ff_bwd, "grouped FF backward",
+          py::arg("dY"), py::arg("tokens"), py::arg("levels"),
+          py::arg("pos"), py::arg("w1"), py::arg("w2"),
+          py::arg("Hpre"), py::arg("Hact"), py::arg("mode"),
+          py::arg("w1t") = c10::nullopt,
+          py::arg("w2t") = c10::nullopt);
     m.def("consensus_fwd", &consensus_fwd, "consensus attention forward");
     m.def("consensus_bwd", &consensus_bwd, "consensus attention backward");
     m.def("level_mix_fwd", &level_mix_fwd, "level mix forward");

@@ -95,13 +95,15 @@ class GlomStepFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, tokens, levels, pos, bw1, bb1, bw2, bb2,
-                tw1, tb1, tw2, tb2, attend_self, mask):
+                tw1, tb1, tw2, tb2, attend_self, mask,
+                bw1t, bw2t, tw1t, tw2t):
         ext = _load_extension()
         out, bhp, bha, thp, tha, probs, rnorm = ext.glom_step_fwd(
             tokens, levels, pos, bw1, bb1, bw2, bb2, tw1, tb1, tw2, tb2,
             attend_self, mask)
         ctx.save_for_backward(tokens, levels, pos, bw1, bw2, tw1, tw2,
-                              bhp, bha, thp, tha, probs, rnorm, mask)
+                              bhp, bha, thp, tha, probs, rnorm, mask,
+                              bw1t, bw2t, tw1t, tw2t)
         ctx.attend_self = attend_self
         return out
 
@@ -109,23 +111,44 @@ class GlomStepFn(torch.autograd.Function):
     def backward(ctx, dnew):
         ext = _load_extension()
         (tokens, levels, pos, bw1, bw2, tw1, tw2, bhp, bha, thp, tha,
-         probs, rnorm, mask) = ctx.saved_tensors
+         probs, rnorm, mask, bw1t, bw2t, tw1t, tw2t) = ctx.saved_tensors
         (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
          dtw1, dtb1, dtw2, dtb2) = ext.glom_step_bwd(
             dnew.contiguous(), tokens, levels, pos, bw1, bw2, tw1, tw2,
-            bhp, bha, thp, tha, probs, rnorm, ctx.attend_self, mask)
+            bhp, bha, thp, tha, probs, rnorm, ctx.attend_self, mask,
+            bw1t, bw2t, tw1t, tw2t)
         return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
-                dtw1, dtb1, dtw2, dtb2, None, None)
+                dtw1, dtb1, dtw2, dtb2, None, None, None, None, None, None)
 
 
-def glom_step(model, tokens, levels, pos, mask):
+def _transposed_weights(model):
+    """Per-group W^T copies for the backward NT GEMMs, computed ONCE per
+    forward (they are loop constants across the T iterations)."""
+    with torch.no_grad():
+        L, d = model.levels, model.dim
+        bw1 = model.bottom_up.net[1].weight[..., 0]
+        bw2 = model.bottom_up.net[3].weight[..., 0]
+        tw1 = model.top_down.net[1].weight[..., 0]
+        tw2 = model.top_down.net[3].weight[..., 0]
+        m4 = bw1.shape[0] // L
+        return (
+            bw1.view(L, m4, d).transpose(1, 2).contiguous(),
+            bw2.view(L, d, m4).transpose(1, 2).contiguous(),
+            tw1.view(L - 1, m4, d).transpose(1, 2).contiguous(),
+            tw2.view(L - 1, d, m4).transpose(1, 2).contiguous(),
+        )
+
+
+def glom_step(model, tokens, levels, pos, mask, wts=None):
     bw = model.bottom_up.net
     tw = model.top_down.net
+    if wts is None:
+        wts = (None, None, None, None)
     return GlomStepFn.apply(
         tokens, levels, pos,
         bw[1].weight[..., 0], bw[1].bias, bw[3].weight[..., 0], bw[3].bias,
         tw[1].weight[..., 0], tw[1].bias, tw[3].weight[..., 0], tw[3].bias,
-        model.attention.attend_self, mask)
+        model.attention.attend_self, mask, *wts)
 
 
 def glom_forward(model, img, iters, levels=None, return_all=False):
@@ -142,9 +165,11 @@ def glom_forward(model, img, iters, levels=None, return_all=False):
     else:
         levels = levels.contiguous()
 
+    wts = _transposed_weights(model) if torch.is_grad_enabled() else None
+
     trajectory = [levels]
     for _ in range(iters):
-        levels = glom_step(model, tokens, levels, pos, mask)
+        levels = glom_step(model, tokens, levels, pos, mask, wts)
         trajectory.append(levels)
 
     if return_all:
