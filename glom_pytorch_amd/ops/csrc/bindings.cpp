@@ -72,6 +72,8 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         launch_gemm_nt_fast3(p, s);
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);
+    else if (tn_fast && p.N % 256 == 0)
+        launch_gemm_tn_fast2(p, s);
     else if (tn_fast)
         launch_gemm_tn_fast(p, s);
     else if (nn_fast)

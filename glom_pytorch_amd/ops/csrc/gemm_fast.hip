@@ -976,3 +976,164 @@ void launch_gemm_nt_fast3(const GemmParams& p, hipStream_t stream) {
     dim3 grid(p.N / BN3, p.M / BM, p.nproblems);
     hipLaunchKernelGGL(gemm_nt_fast3_kernel, grid, dim3(NT3), 0, stream, p);
 }
+
+// ---------------------------------------------------------------- //
+// 128x256-tile TN kernel (8 waves): both operands repack-staged at BK=64,
+// split-K capable. 2x the per-block work of tn_fast for the weight-grad
+// GEMMs; single 68 KiB LDS arena (2 blocks/CU).
+
+__global__ __launch_bounds__(NT3) void gemm_tn_fast2_kernel(GemmParams p) {
+    __shared__ ushort_t smem[128 * EPI2_ROW];    // >= A(128x64)+B(256x64)
+    ushort_t* As = smem;                          // [128][64]
+    ushort_t* Bs = smem + 128 * FBK;              // [256][64]
+
+    int pid = blockIdx.z;
+    int slice = 0, k_begin = 0, k_end = p.K;
+    if (p.splitk > 1) {
+        pid = blockIdx.z % p.nproblems;
+        slice = blockIdx.z / p.nproblems;
+        int per = ((p.K + FBK - 1) / FBK + p.splitk - 1) / p.splitk * FBK;
+        k_begin = slice * per;
+        k_end = min(p.K, k_begin + per);
+    }
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN3;
+    const int m0 = (bid % gridDim.y) * BM;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 2) * 64;
+    const int wn = (wid & 3) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    for (int k0 = k_begin; k0 < k_end; k0 += FBK) {
+        // A: 128 cols -> 16x8=128 slots; B: 256 cols -> 32x8=256 slots
+        if (threadIdx.x < 128)
+            stage_repack(As, Ap, lda, k0, m0, k_end, threadIdx.x);
+        else if (threadIdx.x < 384) {
+            // 256 threads stage B: thread t covers col-block cb = t&31,
+            // m-block mb = t>>5 (same 8x8 micro-tile as stage_repack)
+            int t = threadIdx.x - 128;
+            int cb = t & 31;
+            int mb = t >> 5;
+            union { uint4v v; ushort_t u[8]; } rowv[8];
+#pragma unroll
+            for (int r = 0; r < 8; r++) {
+                int m = k0 + mb * 8 + r;
+                if (m < k_end)
+                    rowv[r].v = *(const uint4v*)(Bp + (long)m * ldb + n0
+                                                 + cb * 8);
+                else
+                    rowv[r].v = 0;
+            }
+#pragma unroll
+            for (int e = 0; e < 8; e++) {
+                union { uint4v v; ushort_t u[8]; } col;
+#pragma unroll
+                for (int r = 0; r < 8; r++) col.u[r] = rowv[r].u[e];
+                int row = cb * 8 + e;
+                int off = (mb * 8) ^ (swz_row(row) << 3);
+                *(uint4v*)&Bs[row * FBK + off] = col.v;
+            }
+        }
+        __syncthreads();
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&As[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bs[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+
+    if (p.splitk > 1) {
+        float* ws = p.ws + ((long)slice * p.nproblems + pid) * p.M * p.N;
+#pragma unroll
+        for (int i16 = 0; i16 < 4; i16++)
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int j = n0 + wn + j16 * 16 + lrow;
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int i = m0 + wm + i16 * 16 + kq * 4 + r;
+                    ws[(long)i * p.N + j] = acc[i16][j16][r];
+                }
+            }
+        return;
+    }
+    // plain epilogue via LDS staging (EPI_NONE only for TN)
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int i16 = 0; i16 < 4; i16++)
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            int li = wm + i16 * 16 + kq * 4 + r;
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int lj = wn + j16 * 16 + lrow;
+                smem[li * EPI2_ROW + lj] =
+                    f2bf(acc[i16][j16][r] * p.alpha);
+            }
+        }
+    __syncthreads();
+    {
+        int t = threadIdx.x;
+        int li = t >> 2;
+        int qt = (t & 3) * 64;
+        long gi = m0 + li;
+        ushort_t* crow = Cp + gi * ldc + n0 + qt;
+        const ushort_t* srow = smem + li * EPI2_ROW + qt;
+#pragma unroll
+        for (int c = 0; c < 8; c++)
+            *(uint4v*)(crow + c * 8) = *(const uint4v*)(srow + c * 8);
+    }
+}
+
+void launch_gemm_tn_fast2(const GemmParams& p, hipStream_t stream) {
+    int sk = p.splitk > 1 ? p.splitk : 1;
+    dim3 grid(p.N / BN3, p.M / BM, p.nproblems * sk);
+    hipLaunchKernelGGL(gemm_tn_fast2_kernel, grid, dim3(NT3), 0, stream, p);
+}
