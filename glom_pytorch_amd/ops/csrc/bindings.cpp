@@ -72,10 +72,8 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         launch_gemm_nt_fast3(p, s);
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);
-    else if (tn_fast && p.N % 256 == 0)
-        launch_gemm_tn_fast2(p, s);
-    else if (tn_fast)
-        launch_gemm_tn_fast(p, s);
+    else if (tn_fast)   // tn_fast2 (128x256) measured slower: 2 blocks/CU
+        launch_gemm_tn_fast(p, s);   // at 68KB LDS lost to 4 blocks/CU here
     else if (nn_fast)
         launch_gemm_nn_fast(p, s);
     else
