@@ -396,34 +396,6 @@ void launch_mix_bwd(const void* dout, void* dmix, void* dtd, long total,
                        (ushort_t*)dtd, total, L, d);
 }
 
-// Column sum over the token axis for bias gradients: in is a batch of P
-// row-major (M x C) matrices (strided), out[p][c] = sum_m in[p][m][c].
-// Split over M with f32 global atomics (out pre-zeroed), coalesced rows.
-__global__ __launch_bounds__(NTHREADS) void k_colsum(
-        const ushort_t* __restrict__ in, float* __restrict__ out,
-        long M, long C, long sin, long ld, int mchunks) {
-    long c = (long)blockIdx.x * NTHREADS + threadIdx.x;
-    if (c >= C) return;
-    int p = blockIdx.z;
-    long step = (M + mchunks - 1) / mchunks;
-    long m0 = (long)blockIdx.y * step;
-    long m1 = m0 + step < M ? m0 + step : M;
-    const ushort_t* base = in + (long)p * sin;
-    float s = 0.f;
-    for (long m = m0; m < m1; m++) s += bf2f(base[m * ld + c]);
-    atomicAdd(&out[p * C + c], s);
-}
-
-void launch_colsum(const void* in, float* out, long P, long M, long C,
-                   long sin, long ld, hipStream_t s) {
-    long mc = (M + 1023) / 1024;
-    int mchunks = (int)(mc < 16 ? mc : 16);
-    if (mchunks < 1) mchunks = 1;
-    dim3 grid((C + NTHREADS - 1) / NTHREADS, mchunks, P);
-    hipLaunchKernelGGL(k_colsum, grid, dim3(NTHREADS), 0, s,
-                       (const ushort_t*)in, out, M, C, sin, ld, mchunks);
-}
-
 // out = a + b + c + d, elementwise bf16 (the four levels-gradient
 // contributions of one GLOM iteration summed in one pass)
 __global__ __launch_bounds__(NTHREADS) void k_add4(
