@@ -1137,3 +1137,355 @@ void launch_gemm_tn_fast2(const GemmParams& p, hipStream_t stream) {
     dim3 grid(p.N / BN3, p.M / BM, p.nproblems * sk);
     hipLaunchKernelGGL(gemm_tn_fast2_kernel, grid, dim3(NT3), 0, stream, p);
 }
+
+// ------- 3-ring variant of nt_fast3 (counted vmcnt, raw barriers) ------
+__global__ __launch_bounds__(NT3) void gemm_nt_fast4_kernel(GemmParams p) {
+    __shared__ ushort_t smem[3 * (128 + 256) * FBK];   // 144 KiB, 3-ring
+    const int abuf = 128 * FBK, bbuf = 256 * FBK, stride = abuf + bbuf;
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN3;
+    const int m0 = (bid % gridDim.y) * BM;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 2) * 64;
+    const int wn = (wid & 3) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    // A: 16 chunks over 8 waves (2 each); B: 32 chunks (4 each)
+    auto stage = [&](int buf, int k0) {
+        ushort_t* Al = smem + buf * stride;
+        ushort_t* Bl = Al + abuf;
+#pragma unroll
+        for (int c = 0; c < 2; c++) {
+            int chunk = wid * 2 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g = Ap + (long)(m0 + row) * lda + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (Al + chunk * 512), 16, 0, 0);
+        }
+#pragma unroll
+        for (int c = 0; c < 4; c++) {
+            int chunk = wid * 4 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g = Bp + (long)(n0 + row) * ldb + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (Bl + chunk * 512), 16, 0, 0);
+        }
+    };
+
+    const int nk = p.K / FBK;
+    // 3-deep ring with counted vmcnt (T3+T4): tile t computes while t+1 is
+    // landed/landing and t+2 streams; ONE raw barrier + ONE counted wait
+    // per K-tile — the vmcnt(0) drain of the 2-buffer loop never happens.
+    stage(0, 0);
+    if (nk > 1) stage(1, FBK);
+    if (nk > 1)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (int kt = 0; kt < nk; kt++) {
+        if (kt + 2 < nk) stage((kt + 2) % 3, (kt + 2) * FBK);
+        const ushort_t* Al = smem + (kt % 3) * stride;
+        const ushort_t* Bl = Al + abuf;
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&Al[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bl[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        if (kt + 1 < nk) {
+            if (kt + 2 < nk)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+    }
+    __syncthreads();   // all waves done computing before epilogue staging
+
+    // ---- epilogue: 128x256 tile via LDS, full-line stores ----
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    ushort_t* out2p = nullptr;
+    if (p.epilogue == EPI_GELU_PAIR)
+        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout;
+
+    float csv[4] = {1.f, 1.f, 1.f, 1.f};
+    float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+    if (csp && p.epilogue != EPI_SMBWD) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+    }
+    if (biasp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+    }
+#pragma unroll
+    for (int i16 = 0; i16 < 4; i16++) {
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            int li = wm + i16 * 16 + kq * 4 + r;           // 0..127
+            long gi = m0 + li;
+            float vv[4];
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++)
+                vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+            if (auxp) {
+                const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                ushort_t av[4];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+                float gx[4], gy[4];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++) gx[j16] = bf2f(av[j16]);
+                gelu_grad_vec<4>(gx, gy);
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++) vv[j16] *= gy[j16];
+            }
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int lj = wn + j16 * 16 + lrow;             // 0..255
+                smem[li * EPI2_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
+            }
+        }
+    }
+    __syncthreads();
+
+    if (p.epilogue == EPI_SOFTMAX || p.epilogue == EPI_SMBWD) {
+        // Full-row fusions: this tile holds complete rows (N == 256), so
+        // the masked row softmax (fwd) / softmax backward (bwd) runs here
+        // instead of a separate kernel + a global-memory round trip.
+        // 4 consecutive threads (one quartet, same wave) own one row.
+        int t = threadIdx.x;
+        int li = t >> 2;
+        int qt = (t & 3) * 64;
+        long gi = m0 + li;
+        const bool* mrow =
+            p.nlmask ? (const bool*)p.nlmask + gi * p.N + n0 + qt : nullptr;
+        ushort_t* srow = smem + li * EPI2_ROW + qt;
+        const float SELF = bf2f(f2bf(-5e-4f));
+        const float NEG = -3.3895314e38f;
+
+        if (p.epilogue == EPI_SOFTMAX) {
+            // pass 1: masks + quartet max
+            float mx = -INFINITY;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++) {
+                    int gj = n0 + qt + c * 8 + e;
+                    float v = bf2f(x.u[e]);
+                    if (p.self_mask && gj == gi) v = SELF;
+                    if (mrow && mrow[c * 8 + e]) v = NEG;
+                    x.u[e] = f2bf(v);
+                    mx = fmaxf(mx, v);
+                }
+                *(uint4v*)(srow + c * 8) = x.v;   // masked values back
+            }
+            mx = fmaxf(mx, __shfl_xor(mx, 1));
+            mx = fmaxf(mx, __shfl_xor(mx, 2));
+            // pass 2: exp + quartet sum (store unnormalized bf16)
+            float sum = 0.f;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++) {
+                    float ex = __expf(bf2f(x.u[e]) - mx);
+                    sum += ex;
+                    x.u[e] = f2bf(ex);
+                }
+                *(uint4v*)(srow + c * 8) = x.v;
+            }
+            sum += __shfl_xor(sum, 1);
+            sum += __shfl_xor(sum, 2);
+            float inv = 1.0f / sum;
+            // pass 3: normalize + store P
+            ushort_t* prow = Cp + gi * ldc + n0 + qt;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++)
+                    x.u[e] = f2bf(bf2f(x.u[e]) * inv);
+                *(uint4v*)(prow + c * 8) = x.v;
+            }
+        } else {
+            // EPI_SMBWD: dS = alpha2 * P * (dP - sum(P*dP)), masked -> 0;
+            // dSr = dS * rnorm[j]. P comes via aux, rnorm via colscale
+            // pointers (phase 1 ran with alpha=1, no colscale).
+            const ushort_t* Prow = (const ushort_t*)p.aux_base
+                + (long)(pid % p.nInner) * p.aux_sin
+                + (long)(pid / p.nInner) * p.aux_sout + gi * p.aux_ld
+                + n0 + qt;
+            const float* rn = (const float*)p.colscale_base
+                + (long)(pid % p.nInner) * p.cs_sin
+                + (long)(pid / p.nInner) * p.cs_sout + n0 + qt;
+            float tsum = 0.f;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } dp, pp;
+                dp.v = *(const uint4v*)(srow + c * 8);
+                pp.v = *(const uint4v*)(Prow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++)
+                    tsum += bf2f(pp.u[e]) * bf2f(dp.u[e]);
+            }
+            tsum += __shfl_xor(tsum, 1);
+            tsum += __shfl_xor(tsum, 2);
+            ushort_t* dsrow = Cp + gi * ldc + n0 + qt;
+            ushort_t* dsr_row = (ushort_t*)p.out2
+                + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout + gi * p.out2_ld
+                + n0 + qt;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } dp, pp, o1, o2;
+                dp.v = *(const uint4v*)(srow + c * 8);
+                pp.v = *(const uint4v*)(Prow + c * 8);
+#pragma unroll
+                for (int e = 0; e < 8; e++) {
+                    int gj = n0 + qt + c * 8 + e;
+                    float v = p.alpha2 * bf2f(pp.u[e])
+                              * (bf2f(dp.u[e]) - tsum);
+                    if (p.self_mask && gj == gi) v = 0.f;
+                    if (mrow && mrow[c * 8 + e]) v = 0.f;
+                    o1.u[e] = f2bf(v);
+                    o2.u[e] = f2bf(v * rn[c * 8 + e]);
+                }
+                *(uint4v*)(dsrow + c * 8) = o1.v;
+                *(uint4v*)(dsr_row + c * 8) = o2.v;
+            }
+        }
+        return;
+    }
+    {
+        int t = threadIdx.x;              // 512 threads: 128 rows x 4 qtrs
+        int li = t >> 2;
+        int qt = (t & 3) * 64;
+        long gi = m0 + li;
+        ushort_t* crow = Cp + gi * ldc + n0 + qt;
+        ushort_t* orow = out2p ? out2p + gi * p.out2_ld + n0 + qt : nullptr;
+        const ushort_t* srow = smem + li * EPI2_ROW + qt;
+#pragma unroll
+        for (int c = 0; c < 8; c++)
+            *(uint4v*)(crow + c * 8) = *(const uint4v*)(srow + c * 8);
+        if (orow) {
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x, g;
+                x.v = *(const uint4v*)(srow + c * 8);
+                if (p.epilogue == 10) {
+                    g.v = x.v;
+                } else if (p.epilogue == 12) {      // debug: poly, no trans
+#pragma unroll
+                    for (int e = 0; e < 8; e++) {
+                        float xv = bf2f(x.u[e]);
+                        g.u[e] = f2bf(xv * (0.5f + 0.1f * xv));
+                    }
+                } else {
+                    float xin[8], yv[8];
+#pragma unroll
+                    for (int e = 0; e < 8; e++) xin[e] = bf2f(x.u[e]);
+                    gelu_f_vec<8>(xin, yv);
+#pragma unroll
+                    for (int e = 0; e < 8; e++) g.u[e] = f2bf(yv[e]);
+                }
+                *(uint4v*)(orow + c * 8) = g.v;
+            }
+        }
+    }
+    if (p.colsum_out) {
+        // fused bias-grad: this tile's column sums (C values still live in
+        // the LDS staging image) accumulated into the f32 output
+        float* outp = p.colsum_out + (long)(pid % p.nInner) * p.colsum_sin;
+        int c = threadIdx.x;
+        if (c < BN3) {
+            float ssum = 0.f;
+            for (int r = 0; r < 128; r++)
+                ssum += bf2f(smem[r * EPI2_ROW + c]);
+            atomicAdd(&outp[n0 + c], ssum);
+        }
+    }
+}
+
+
+void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN3, p.M / BM, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast4_kernel, grid, dim3(NT3), 0, stream, p);
+}
