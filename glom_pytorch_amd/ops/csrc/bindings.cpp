@@ -68,10 +68,8 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
     const bool nt3 = nt_fast && p.N % 256 == 0 && p.N >= 1024;
     if (nt2)
         launch_gemm_nt_fast2(p, s);
-    else if (nt3 && getenv("GLOM_NT4"))
-        launch_gemm_nt_fast4(p, s);
     else if (nt3)
-        launch_gemm_nt_fast3(p, s);
+        launch_gemm_nt_fast4(p, s);   // 3-ring counted-vmcnt variant
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);
     else if (tn_fast)   // tn_fast2 (128x256) measured slower: 2 blocks/CU
