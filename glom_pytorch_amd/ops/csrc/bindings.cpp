@@ -70,6 +70,8 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         launch_gemm_nt_fast2(p, s);
     else if (nt3)
         launch_gemm_nt_fast4(p, s);   // 3-ring counted-vmcnt variant
+    else if (nt_fast && getenv("GLOM_NT5"))
+        launch_gemm_nt_fast5(p, s);
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);
     else if (tn_fast)   // tn_fast2 (128x256) measured slower: 2 blocks/CU
@@ -336,7 +338,7 @@ std::vector<torch::Tensor> consensus_fwd(
             p.self_mask = attend_self ? 0 : 1;
             p.nlmask = mask;
             p.splitk = 1;
-            launch_gemm_nt_fast3(p, s);
+            launch_gemm_nt_fast4(p, s);
             check_launch();
         } else {
             run_gemm(p, s, opts, lds_ok);
@@ -398,7 +400,7 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
         p.nlmask = mask;
         p.alpha2 = (float)std::pow((double)d, -0.5);
         p.splitk = 1;
-        launch_gemm_nt_fast3(p, s);
+        launch_gemm_nt_fast4(p, s);
         check_launch();
     } else {
         dP = torch::empty({B, L, N, N}, opts);

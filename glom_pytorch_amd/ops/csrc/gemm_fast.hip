@@ -1489,3 +1489,102 @@ void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream) {
     dim3 grid(p.N / BN3, p.M / BM, p.nproblems);
     hipLaunchKernelGGL(gemm_nt_fast4_kernel, grid, dim3(NT3), 0, stream, p);
 }
+\n// ------- 3-ring variant of nt_fast (128^2, 1 block/CU) -------\n__global__ __launch_bounds__(NTHREADS) void gemm_nt_fast5_kernel(GemmParams p) {
+    // ONE shared array: a second __shared__ object makes hipcc emit
+    // s_waitcnt vmcnt(0) before the first ds_read of every k-step of a
+    // glds pipeline, draining the prefetch (guide §5 ".s-level traps" (a)).
+    __shared__ ushort_t smem[6 * BM * FBK];   // 3-ring x (A|B)
+    ushort_t* As0 = smem;
+    ushort_t* Bs0 = smem + 3 * BM * FBK;
+
+    const int pid = blockIdx.z;
+    // XCD-aware block remap (T1), column-major: each XCD die owns a
+    // contiguous run of N-columns (all M-tiles of a few n-tiles), so the
+    // shared B panel of a column stays resident in that XCD's private L2
+    // while the M sweep streams A.
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN;   // column-major: n outer
+    const int m0 = (bid % gridDim.y) * BM;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 1) * 64;
+    const int wn = (wid & 1) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    const int nk = p.K / FBK;
+    stage_glds(As0, Ap, lda, m0, 0, wid, lane);
+    stage_glds(Bs0, Bp, ldb, n0, 0, wid, lane);
+    if (nk > 1) {
+        stage_glds(As0 + BM * FBK, Ap, lda, m0, FBK, wid, lane);
+        stage_glds(Bs0 + BM * FBK, Bp, ldb, n0, FBK, wid, lane);
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    for (int kt = 0; kt < nk; kt++) {
+        int cur = kt % 3;
+        if (kt + 2 < nk) {
+            stage_glds(As0 + ((kt + 2) % 3) * BM * FBK, Ap, lda, m0,
+                       (kt + 2) * FBK, wid, lane);
+            stage_glds(Bs0 + ((kt + 2) % 3) * BM * FBK, Bp, ldb, n0,
+                       (kt + 2) * FBK, wid, lane);
+        }
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&As0[cur * BM * FBK + row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bs0[cur * BN * FBK + row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        if (kt + 1 < nk) {
+            if (kt + 2 < nk)
+                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+    }
+    __syncthreads();
+    gemm_epilogue_lds(p, pid, m0, n0, wm, wn, lrow, kq, acc, smem);
+}
+
+
+void launch_gemm_nt_fast5(const GemmParams& p, hipStream_t stream) {
+    dim3 grid((p.N + BN - 1) / BN, (p.M + BM - 1) / BM, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast5_kernel, grid, dim3(NTHREADS), 0,
+                       stream, p);
+}
