@@ -1489,7 +1489,9 @@ void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream) {
     dim3 grid(p.N / BN3, p.M / BM, p.nproblems);
     hipLaunchKernelGGL(gemm_nt_fast4_kernel, grid, dim3(NT3), 0, stream, p);
 }
-\n// ------- 3-ring variant of nt_fast (128^2, 1 block/CU) -------\n__global__ __launch_bounds__(NTHREADS) void gemm_nt_fast5_kernel(GemmParams p) {
+
+// ------- 3-ring variant of nt_fast (128^2, 1 block/CU) -------
+__global__ __launch_bounds__(NTHREADS) void gemm_nt_fast5_kernel(GemmParams p) {
     // ONE shared array: a second __shared__ object makes hipcc emit
     // s_waitcnt vmcnt(0) before the first ds_read of every k-step of a
     // glds pipeline, draining the prefetch (guide §5 ".s-level traps" (a)).
