@@ -38,6 +38,9 @@ def parse_args():
     p.add_argument("--mode", choices=["train", "infer"], default="train",
                    help="train = denoising fwd+bwd+AdamW (headline); "
                         "infer = no-grad forward under hipGraph replay")
+    p.add_argument("--profile", action="store_true",
+                   help="emit roctx ranges around each timed step (pair "
+                        "with rocprofv3 --marker-trace)")
     return p.parse_args()
 
 
@@ -92,8 +95,14 @@ def main():
 
     sync_all()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        loss = step(img, args.iters)
+    if args.profile:
+        from glom_pytorch_amd.utils.profiling import trace_range
+        for i in range(args.steps):
+            with trace_range(f"bench/step{i}"):
+                loss = step(img, args.iters)
+    else:
+        for _ in range(args.steps):
+            loss = step(img, args.iters)
     sync_all()
     elapsed = time.perf_counter() - t0
 

@@ -152,8 +152,10 @@ def glom_step(model, tokens, levels, pos, mask, wts=None):
 
 
 def glom_forward(model, img, iters, levels=None, return_all=False):
+    from glom_pytorch_amd.utils.profiling import trace_range
     b = img.shape[0]
-    tokens = model.image_to_tokens(img)          # K1: once per forward
+    with trace_range("glom/patch_embed"):
+        tokens = model.image_to_tokens(img)      # K1: once per forward
     n = tokens.shape[1]
     pos = model.pos_emb.weight
     mask = (model.attention.non_local_mask
@@ -168,9 +170,11 @@ def glom_forward(model, img, iters, levels=None, return_all=False):
     wts = _transposed_weights(model) if torch.is_grad_enabled() else None
 
     trajectory = [levels]
-    for _ in range(iters):
-        levels = glom_step(model, tokens, levels, pos, mask, wts)
-        trajectory.append(levels)
+    with trace_range(f"glom/iterate x{iters}"):
+        for t in range(iters):
+            with trace_range(f"glom/step{t}"):
+                levels = glom_step(model, tokens, levels, pos, mask, wts)
+            trajectory.append(levels)
 
     if return_all:
         return torch.stack(trajectory)

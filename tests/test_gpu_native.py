@@ -258,3 +258,30 @@ def test_consensus_fused_softmax_n256(attend_self, radius):
     (dref,) = torch.autograd.grad(ref, x, g.float())
     assert _rel_err(out, ref) < 1e-2
     assert _rel_err(dlev, dref) < 3e-2
+
+
+def test_forward_determinism_race_smoke():
+    """Two identical forwards must agree bitwise: the forward path has no
+    atomics or cross-block races by construction (SURVEY.md §5 race
+    detection strategy)."""
+    torch.manual_seed(0)
+    m = Glom(dim=512, levels=6, image_size=224, patch_size=14)
+    m = m.to(DEV, torch.bfloat16)
+    img = torch.randn(2, 3, 224, 224, device=DEV, dtype=torch.bfloat16)
+    with torch.no_grad():
+        a = m(img, iters=4)
+        b = m(img, iters=4)
+    assert torch.equal(a, b)
+
+
+def test_denoising_trainer_gpu_step():
+    from glom_pytorch_amd.parallel.trainer import DenoisingTrainer
+    torch.manual_seed(0)
+    m = Glom(**CFG).to(DEV, torch.bfloat16)
+    tr = DenoisingTrainer(m, noise_std=0.5, decode_step=2)
+    img = torch.randn(2, 3, 32, 32, device=DEV, dtype=torch.bfloat16)
+    l1 = tr.step(img, iters=3)
+    l2 = tr.step(img, iters=3)
+    assert l1 > 0 and l2 > 0
+    for n, p in m.named_parameters():
+        assert torch.isfinite(p.float()).all(), n
