@@ -321,6 +321,9 @@ std::vector<torch::Tensor> consensus_fwd(
 
     auto probs = torch::empty({B, L, N, N}, opts);
     const bool fused_sm = (N == 256) && (d % 64 == 0) && lds_ok;
+    const bool fuse_av = fused_sm && !getenv("GLOM_NO_FUSE_AV");
+    torch::Tensor out_av;
+    if (fuse_av) out_av = torch::empty({B, N, L, d}, opts);
     // scores[i,j] = (q_i . k_j) * rnorm_j * d^-0.5, then masked row softmax
     {
         GemmParams p = base_params(N, N, d, LAYOUT_NT, P, L,
@@ -338,6 +341,14 @@ std::vector<torch::Tensor> consensus_fwd(
             p.self_mask = attend_self ? 0 : 1;
             p.nlmask = mask;
             p.splitk = 1;
+            if (fuse_av) {
+                // AV product fused too: O computed from the LDS-resident P
+                p.out2 = out_av.data_ptr();
+                p.out2_sin = d; p.out2_sout = N * L * d; p.out2_ld = L * d;
+                p.aux_base = levels.data_ptr();
+                p.aux_sin = d; p.aux_sout = N * L * d; p.aux_ld = L * d;
+                p.npatch = (int)d;
+            }
             launch_gemm_nt_fast4(p, s);
             check_launch();
         } else {
@@ -348,6 +359,8 @@ std::vector<torch::Tensor> consensus_fwd(
         }
     }
     // out[i,:] = sum_j P[i,j] * levels[j,:]
+    if (fuse_av)
+        return {out_av, probs, rnorm};
     auto out = torch::empty({B, N, L, d}, opts);
     {
         GemmParams p = base_params(N, d, N, LAYOUT_NN, P, L, 1.0f);

@@ -1374,7 +1374,8 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast4_kernel(GemmParams p) {
             sum += __shfl_xor(sum, 1);
             sum += __shfl_xor(sum, 2);
             float inv = 1.0f / sum;
-            // pass 3: normalize + store P
+            // pass 3: normalize; P goes to global AND stays in LDS for the
+            // fused AV product below
             ushort_t* prow = Cp + gi * ldc + n0 + qt;
 #pragma unroll
             for (int c = 0; c < 8; c++) {
@@ -1384,6 +1385,101 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast4_kernel(GemmParams p) {
                 for (int e = 0; e < 8; e++)
                     x.u[e] = f2bf(bf2f(x.u[e]) * inv);
                 *(uint4v*)(prow + c * 8) = x.v;
+                *(uint4v*)(srow + c * 8) = x.v;
+            }
+            if (p.out2) {
+                // ---- fused AV: O[i,:] = sum_j P[i,j] * V[j,:] ----
+                // P: LDS [128][EPI2_ROW]; V: global rows via aux fields;
+                // O: out2 fields; d passed in p.npatch. K = N = 256.
+                __syncthreads();
+                const ushort_t* Vp = (const ushort_t*)p.aux_base
+                    + (long)(pid % p.nInner) * p.aux_sin
+                    + (long)(pid / p.nInner) * p.aux_sout;
+                const long ldv = p.aux_ld;
+                ushort_t* Op = (ushort_t*)p.out2
+                    + (long)(pid % p.nInner) * p.out2_sin
+                    + (long)(pid / p.nInner) * p.out2_sout;
+                const int dtot = p.npatch;
+                ushort_t* Vt = smem + 128 * EPI2_ROW;   // [256][FBK]
+                const int wrow = (wid >> 2) * 64;
+                const int wcol64 = (wid & 3) * 64;
+                for (int dc = 0; dc < dtot; dc += 256) {
+                    f32x4 acc2[4][4] = {};
+                    for (int jt = 0; jt < 4; jt++) {
+                        // transpose-stage V rows jt*64.. cols dc..dc+255
+                        if (threadIdx.x < 256) {
+                            int tt = threadIdx.x;
+                            int cb = tt & 31;
+                            int mb = tt >> 5;
+                            union { uint4v v; ushort_t u[8]; } rv[8];
+#pragma unroll
+                            for (int r = 0; r < 8; r++) {
+                                int j = jt * 64 + mb * 8 + r;
+                                int gc = dc + cb * 8;
+                                if (gc + 7 < dtot)
+                                    rv[r].v = *(const uint4v*)(Vp
+                                        + (long)j * ldv + gc);
+                                else
+                                    rv[r].v = 0;
+                            }
+#pragma unroll
+                            for (int e = 0; e < 8; e++) {
+                                union { uint4v v; ushort_t u[8]; } col;
+#pragma unroll
+                                for (int r = 0; r < 8; r++)
+                                    col.u[r] = rv[r].u[e];
+                                int row = cb * 8 + e;
+                                int off = (mb * 8) ^ (swz_row(row) << 3);
+                                *(uint4v*)&Vt[row * FBK + off] = col.v;
+                            }
+                        }
+                        __syncthreads();
+                        short8 paf[2][4], vbf[2][4];
+#pragma unroll
+                        for (int ss = 0; ss < 2; ss++) {
+#pragma unroll
+                            for (int i = 0; i < 4; i++) {
+                                int row = wrow + i * 16 + lrow;
+                                int koff = jt * 64 + ss * 32 + kq * 8;
+                                paf[ss][i] = *(const short8*)&smem[
+                                    row * EPI2_ROW + koff];
+                            }
+#pragma unroll
+                            for (int j = 0; j < 4; j++) {
+                                int row = wcol64 + j * 16 + lrow;
+                                int off = (ss * 32 + kq * 8)
+                                          ^ (swz_row(row) << 3);
+                                vbf[ss][j] = *(const short8*)&Vt[
+                                    row * FBK + off];
+                            }
+                        }
+#pragma unroll
+                        for (int ss = 0; ss < 2; ss++)
+#pragma unroll
+                            for (int i = 0; i < 4; i++)
+#pragma unroll
+                                for (int j = 0; j < 4; j++)
+                                    acc2[i][j] =
+                                        __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                            paf[ss][i], vbf[ss][j],
+                                            acc2[i][j], 0, 0, 0);
+                        __syncthreads();
+                    }
+                    // store O block: rows m0+wrow.., cols dc+wcol64..
+#pragma unroll
+                    for (int i16 = 0; i16 < 4; i16++)
+#pragma unroll
+                        for (int r = 0; r < 4; r++) {
+                            long oi = m0 + wrow + i16 * 16 + kq * 4 + r;
+                            ushort_t* orow2 = Op + oi * p.out2_ld;
+#pragma unroll
+                            for (int j16 = 0; j16 < 4; j16++) {
+                                int oc = dc + wcol64 + j16 * 16 + lrow;
+                                if (oc < dtot)
+                                    orow2[oc] = f2bf(acc2[i16][j16][r]);
+                            }
+                        }
+                }
             }
         } else {
             // EPI_SMBWD: dS = alpha2 * P * (dP - sum(P*dP)), masked -> 0;
