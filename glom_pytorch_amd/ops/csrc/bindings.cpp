@@ -74,8 +74,10 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         launch_gemm_nt_fast5(p, s);
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);
-    else if (tn_fast)   // tn_fast2 (128x256) measured slower: 2 blocks/CU
-        launch_gemm_tn_fast(p, s);   // at 68KB LDS lost to 4 blocks/CU here
+    else if (tn_fast && p.splitk > 1)
+        launch_gemm_tn_sk(p, s);     // 32KB arena: 4 blocks/CU for the
+    else if (tn_fast)                // weight-grad split-K launches
+        launch_gemm_tn_fast(p, s);
     else if (nn_fast)
         launch_gemm_nn_fast(p, s);
     else

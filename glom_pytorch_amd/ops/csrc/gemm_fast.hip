@@ -1686,3 +1686,110 @@ void launch_gemm_nt_fast5(const GemmParams& p, hipStream_t stream) {
     hipLaunchKernelGGL(gemm_nt_fast5_kernel, grid, dim3(NTHREADS), 0,
                        stream, p);
 }
+
+// ------- split-K-only TN variant (32 KiB arena, 4 blocks/CU) -------
+__global__ __launch_bounds__(NTHREADS) void gemm_tn_sk_kernel(GemmParams p) {
+    __shared__ ushort_t smem[2 * BM * FBK];   // 32 KiB: 4 blocks/CU
+    ushort_t* As = smem;
+    ushort_t* Bs = smem + BM * FBK;
+
+    int pid = blockIdx.z;
+    int slice = 0, k_begin = 0, k_end = p.K;
+    if (p.splitk > 1) {
+        pid = blockIdx.z % p.nproblems;
+        slice = blockIdx.z / p.nproblems;
+        int per = ((p.K + FBK - 1) / FBK + p.splitk - 1) / p.splitk * FBK;
+        k_begin = slice * per;
+        k_end = min(p.K, k_begin + per);
+    }
+    // XCD-aware block remap (T1), column-major: each XCD die owns a
+    // contiguous run of N-columns (all M-tiles of a few n-tiles), so the
+    // shared B panel of a column stays resident in that XCD's private L2
+    // while the M sweep streams A.
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN;   // column-major: n outer
+    const int m0 = (bid % gridDim.y) * BM;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 1) * 64;
+    const int wn = (wid & 1) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[4][4] = {};
+
+    for (int k0 = k_begin; k0 < k_end; k0 += FBK) {
+        if (threadIdx.x < 128)
+            stage_repack(As, Ap, lda, k0, m0, k_end, threadIdx.x);
+        else
+            stage_repack(Bs, Bp, ldb, k0, n0, k_end, threadIdx.x - 128);
+        __syncthreads();
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&As[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bs[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+        __syncthreads();
+    }
+
+    if (p.splitk > 1) {
+        float* ws = p.ws + ((long)slice * p.nproblems + pid) * p.M * p.N;
+#pragma unroll
+        for (int i16 = 0; i16 < 4; i16++)
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                int j = n0 + wn + j16 * 16 + lrow;
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int i = m0 + wm + i16 * 16 + kq * 4 + r;
+                    ws[(long)i * p.N + j] = acc[i16][j16][r];
+                }
+            }
+        return;
+    }
+    // non-split-K falls back to the generic-epilogue variant (dispatch
+    // guarantees splitk > 1 here)
+}
+
+void launch_gemm_nt_fast(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN, p.M / BM, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast_kernel, grid, dim3(NTHREADS), 0, stream,
+                       p);
+}
+
+
+void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN, p.M / BM, p.nproblems * p.splitk);
+    hipLaunchKernelGGL(gemm_tn_sk_kernel, grid, dim3(NTHREADS), 0, stream, p);
+}
