@@ -1782,11 +1782,7 @@ __global__ __launch_bounds__(NTHREADS) void gemm_tn_sk_kernel(GemmParams p) {
     // guarantees splitk > 1 here)
 }
 
-void launch_gemm_nt_fast(const GemmParams& p, hipStream_t stream) {
-    dim3 grid(p.N / BN, p.M / BM, p.nproblems);
-    hipLaunchKernelGGL(gemm_nt_fast_kernel, grid, dim3(NTHREADS), 0, stream,
-                       p);
-}
+
 
 
 void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
