@@ -55,7 +55,8 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
     if distributed:
-        dist.init_process_group("nccl")
+        from glom_pytorch_amd.parallel.failure import init_distributed
+        init_distributed(timeout_s=300)
     torch.cuda.set_device(local_rank)
     dev = torch.device("cuda", local_rank)
     torch.manual_seed(1234 + rank)

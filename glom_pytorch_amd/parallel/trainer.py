@@ -17,6 +17,7 @@ from einops.layers.torch import Rearrange
 from torch import nn
 
 from glom_pytorch_amd.parallel.ddp import BucketedDDP
+from glom_pytorch_amd.parallel.failure import Heartbeat
 
 
 class DenoisingDecoder(nn.Sequential):
@@ -43,10 +44,12 @@ class DenoisingTrainer:
             model.dim, model.image_size, model.patch_size
         ).to(p.device, p.dtype)
         self.distributed = distributed and dist.is_initialized()
-        self.ddp_model = self.ddp_dec = None
+        self.ddp_model = self.ddp_dec = self.heartbeat = None
         if self.distributed:
             self.ddp_model = BucketedDDP(model, bucket_bytes)
             self.ddp_dec = BucketedDDP(self.decoder, bucket_bytes)
+            # rank-liveness probe: a hung peer becomes a clean abort
+            self.heartbeat = Heartbeat(every_steps=50)
         self.opt = torch.optim.AdamW(
             list(model.parameters()) + list(self.decoder.parameters()),
             lr=lr, foreach=True)
@@ -66,6 +69,7 @@ class DenoisingTrainer:
         if self.distributed:
             self.ddp_model.finalize()
             self.ddp_dec.finalize()
+            self.heartbeat.tick()
         self.opt.step()
         self.step_idx += 1
         return loss.item()
