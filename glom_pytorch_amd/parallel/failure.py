@@ -28,8 +28,10 @@ class Heartbeat:
 
     def __init__(self, every_steps: int = 50, device=None):
         self.every = every_steps
-        self.device = device or torch.device(
-            "cuda", torch.cuda.current_device())
+        if device is None:
+            device = (torch.device("cuda", torch.cuda.current_device())
+                      if torch.cuda.is_available() else torch.device("cpu"))
+        self.device = device
         self._buf = torch.ones(1, device=self.device)
         self._step = 0
 
