@@ -85,3 +85,4 @@ void launch_gemm_tn_fast2(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast5(const GemmParams& p, hipStream_t stream);
 void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream);
+void launch_gemm_nt_fast6(const GemmParams& p, hipStream_t stream);

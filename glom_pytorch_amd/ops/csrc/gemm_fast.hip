@@ -1789,3 +1789,232 @@ void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
     dim3 grid(p.N / BN, p.M / BM, p.nproblems * p.splitk);
     hipLaunchKernelGGL(gemm_tn_sk_kernel, grid, dim3(NTHREADS), 0, stream, p);
 }
+
+
+// ---------------------------------------------------------------- //
+// 256x256-tile NT kernel, BK=32, 3-deep counted-vmcnt ring: the only
+// square (intensity-128 flops/byte) tile whose triple buffer fits LDS
+// (3 x 32 KiB). K-rows are stored PAIRED (two 64B k-rows per 128B LDS
+// row, linear for glds) so ds_read_b128 stays in the 64-bank regime;
+// the granule swizzle spreads the 16-lane read groups conflict-free.
+
+#define BK32 32
+
+__global__ __launch_bounds__(NT2) void gemm_nt_fast6_kernel(GemmParams p) {
+    __shared__ ushort_t smem[3 * 2 * BM2 * BK32];   // 96 KiB
+    const int tsz = BM2 * BK32;                     // elems per operand tile
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN2;
+    const int m0 = (bid % gridDim.y) * BM2;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 2) * 128;
+    const int wn = (wid & 3) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    f32x4 acc[8][4] = {};
+
+    // stage one K-step (256 rows x 32 k per operand = 16 chunks, 2 per
+    // wave per operand). Paired-row image: k-rows 2R,2R+1 share LDS row R.
+    auto stage = [&](int buf, int k0) {
+#pragma unroll
+        for (int op = 0; op < 2; op++) {
+            const ushort_t* src = op ? Bp : Ap;
+            long ld = op ? ldb : lda;
+            int base0 = op ? n0 : m0;
+            ushort_t* lds = smem + (buf * 2 + op) * tsz;
+#pragma unroll
+            for (int c = 0; c < 2; c++) {
+                int chunk = wid * 2 + c;
+                int R = chunk * 8 + (lane >> 3);        // image row (128B)
+                int gg = (lane & 7) ^ swz_row(R);       // swizzled granule
+                int row = R * 2 + (gg >> 2);            // global tile row
+                int kpart = (gg & 3) * 8;
+                const ushort_t* gaddr =
+                    src + (long)(base0 + row) * ld + k0 + kpart;
+                ushort_t* laddr = lds + chunk * 512;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) unsigned int*)
+                        gaddr,
+                    (__attribute__((address_space(3))) unsigned int*)laddr,
+                    16, 0, 0);
+            }
+        }
+    };
+
+    const int nk = p.K / BK32;
+    stage(0, 0);
+    if (nk > 1) stage(1, BK32);
+    if (nk > 1)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (int kt = 0; kt < nk; kt++) {
+        if (kt + 2 < nk) stage((kt + 2) % 3, (kt + 2) * BK32);
+        const ushort_t* Al = smem + ((kt % 3) * 2 + 0) * tsz;
+        const ushort_t* Bl = smem + ((kt % 3) * 2 + 1) * tsz;
+        short8 bfr[4];
+#pragma unroll
+        for (int j = 0; j < 4; j++) {
+            int row = wn + j * 16 + lrow;
+            int R = row >> 1;
+            int gg = ((row & 1) * 4 + kq) ^ swz_row(R);
+            bfr[j] = *(const short8*)&Bl[R * 64 + gg * 8];
+        }
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+            int row = wm + i * 16 + lrow;
+            int R = row >> 1;
+            int gg = ((row & 1) * 4 + kq) ^ swz_row(R);
+            short8 af = *(const short8*)&Al[R * 64 + gg * 8];
+#pragma unroll
+            for (int j = 0; j < 4; j++)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af, bfr[j], acc[i][j], 0, 0, 0);
+        }
+        if (kt + 1 < nk) {
+            if (kt + 2 < nk)
+                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+    }
+    __syncthreads();
+
+    // ---- epilogue: two half-tiles (rows 0-127, 128-255) via LDS ----
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    ushort_t* out2p = nullptr;
+    if (p.epilogue == EPI_GELU_PAIR)
+        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
+                + (long)(pid / p.nInner) * p.out2_sout;
+
+#pragma unroll
+    for (int half = 0; half < 2; half++) {
+        if (wm == half * 128) {
+            float csv[4] = {1.f, 1.f, 1.f, 1.f};
+            float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+            if (csp) {
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+            }
+            if (biasp) {
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+            }
+#pragma unroll
+            for (int i16 = 0; i16 < 8; i16++) {
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    int li = i16 * 16 + kq * 4 + r;        // 0..127
+                    long gi = m0 + half * 128 + li;
+                    float vv[4];
+#pragma unroll
+                    for (int j16 = 0; j16 < 4; j16++)
+                        vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+                    if (auxp) {
+                        const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                        ushort_t av[4];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            vv[j16] *= gelu_grad_f(bf2f(av[j16]));
+                    }
+#pragma unroll
+                    for (int j16 = 0; j16 < 4; j16++) {
+                        int lj = wn + j16 * 16 + lrow;     // 0..255
+                        smem[li * EPI2_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
+                    }
+                }
+            }
+        }
+        __syncthreads();
+        {
+            int t = threadIdx.x;          // 512 threads: 128 rows x 4 qtrs
+            int li = t >> 2;
+            int qt = (t & 3) * 64;
+            long gi = m0 + half * 128 + li;
+            ushort_t* crow = Cp + gi * ldc + n0 + qt;
+            ushort_t* orow =
+                out2p ? out2p + gi * p.out2_ld + n0 + qt : nullptr;
+            const ushort_t* srow = smem + li * EPI2_ROW + qt;
+#pragma unroll
+            for (int c = 0; c < 8; c++) {
+                union { uint4v v; ushort_t u[8]; } x;
+                x.v = *(const uint4v*)(srow + c * 8);
+                *(uint4v*)(crow + c * 8) = x.v;
+                if (orow) {
+                    union { uint4v v; ushort_t u[8]; } g;
+#pragma unroll
+                    for (int e = 0; e < 8; e++)
+                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
+                    *(uint4v*)(orow + c * 8) = g.v;
+                }
+            }
+        }
+        if (p.colsum_out) {
+            float* outp = p.colsum_out
+                + (long)(pid % p.nInner) * p.colsum_sin;
+            int c = threadIdx.x;
+            if (c < BN2) {
+                float ssum = 0.f;
+                for (int r = 0; r < 128; r++)
+                    ssum += bf2f(smem[r * EPI2_ROW + c]);
+                atomicAdd(&outp[n0 + c], ssum);
+            }
+        }
+        __syncthreads();
+    }
+}
+
+
+void launch_gemm_nt_fast6(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN2, p.M / BM2, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast6_kernel, grid, dim3(NT2), 0, stream, p);
+}
