@@ -64,16 +64,9 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
             p.ws = ws.data_ptr<float>();
         }
     }
-    const bool nt2 = false;   // 256^2 phase kernel: kept for later tuning
     const bool nt3 = nt_fast && p.N % 256 == 0 && p.N >= 1024;
-    if (nt2)
-        launch_gemm_nt_fast2(p, s);
-    else if (nt3 && p.M % 256 == 0 && p.N % 256 == 0 && getenv("GLOM_NT6"))
-        launch_gemm_nt_fast6(p, s);   // 256^2 BK=32 ring (A/B gate)
-    else if (nt3)
+    if (nt3)
         launch_gemm_nt_fast4(p, s);   // 3-ring counted-vmcnt variant
-    else if (nt_fast && getenv("GLOM_NT5"))
-        launch_gemm_nt_fast5(p, s);
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);
     else if (tn_fast && p.splitk > 1)

@@ -79,10 +79,6 @@ void launch_gemm_finish(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast(const GemmParams& p, hipStream_t stream);
 void launch_gemm_tn_fast(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nn_fast(const GemmParams& p, hipStream_t stream);
-void launch_gemm_nt_fast2(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast3(const GemmParams& p, hipStream_t stream);
-void launch_gemm_tn_fast2(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream);
-void launch_gemm_nt_fast5(const GemmParams& p, hipStream_t stream);
 void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream);
-void launch_gemm_nt_fast6(const GemmParams& p, hipStream_t stream);

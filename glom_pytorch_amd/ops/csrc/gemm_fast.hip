@@ -369,264 +369,7 @@ void launch_gemm_nn_fast(const GemmParams& p, hipStream_t stream) {
                        p);
 }
 
-// ---------------------------------------------------------------- //
-// 256x256-tile NT kernel: 8 waves (512 threads), per-wave 128x64 output,
-// BK=64 double-buffered glds staging (128 KiB LDS, 1 block/CU, 2 waves/
-// SIMD). 4x the work per block of the 128^2 kernel at 2x the arithmetic
-// intensity (128 flops/byte) — amortizes the per-block prologue/epilogue
-// that dominates the short-K (K=512) GLOM shapes.
 
-#define NT2 512
-#define BM2 256
-#define BN2 256
-
-// stage one 256-row x 64-col operand tile: 64 KiB = 32 DMA chunks, 4 per wave
-__device__ __forceinline__ void stage_glds256(
-        ushort_t* lds, const ushort_t* src, long ld, int r0, int k0,
-        int wid, int lane) {
-#pragma unroll
-    for (int c = 0; c < 4; c++) {
-        int chunk = wid * 4 + c;
-        int row = chunk * 8 + (lane >> 3);
-        int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
-        const ushort_t* gaddr = src + (long)(r0 + row) * ld + k0 + swz8;
-        ushort_t* laddr = lds + chunk * 512;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int*)gaddr,
-            (__attribute__((address_space(3))) unsigned int*)laddr, 16, 0, 0);
-    }
-}
-
-#define EPI2_ROW 264   // 256 cols + 8 pad
-
-__global__ __launch_bounds__(NT2) void gemm_nt_fast2_kernel(GemmParams p) {
-    // 8 half-tile slots (16 KiB each = 128 rows x 64 k bf16): a 2-deep
-    // double buffer at HALF-tile granularity. Prefetch runs 3-7 halves
-    // ahead with ONE counted s_waitcnt vmcnt(6) per K-tile (2 glds/wave
-    // per half x 3 halves in flight) and raw barriers — the 2-phase
-    // drain-every-tile stall never happens (guide T3+T4).
-    __shared__ ushort_t smem[8 * 128 * FBK];
-
-    const int pid = blockIdx.z;
-    int nwg = gridDim.x * gridDim.y;
-    int bid = blockIdx.y * gridDim.x + blockIdx.x;
-    {
-        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
-        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
-    }
-    const int n0 = (bid / gridDim.y) * BN2;
-    const int m0 = (bid % gridDim.y) * BM2;
-
-    const ushort_t* Ap;
-    const ushort_t* Bp;
-    long lda, ldb;
-    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
-    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
-
-    const int wid = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int wm = (wid >> 2) * 128;
-    const int wn = (wid & 3) * 64;
-    const int lrow = lane & 15;
-    const int kq = lane >> 4;
-
-    f32x4 acc[8][4] = {};
-
-    const int nk = p.K / FBK;
-    const int total_halves = nk * 4;
-
-    // stage one 128-row half-tile (2 glds per wave): h = 0/1 -> A rows
-    // 0-127/128-255, 2/3 -> B rows 0-127/128-255 of this block's tiles
-    auto stage_half = [&](int half_id) {
-        int tile = half_id >> 2;
-        int h = half_id & 3;
-        const ushort_t* src = (h & 2) ? Bp : Ap;
-        long ld = (h & 2) ? ldb : lda;
-        int base0 = ((h & 2) ? n0 : m0) + (h & 1) * 128;
-        ushort_t* lds = smem + (((tile & 1) << 2) | h) * (128 * FBK);
-        int k0 = tile * FBK;
-#pragma unroll
-        for (int c = 0; c < 2; c++) {
-            int chunk = wid * 2 + c;
-            int row = chunk * 8 + (lane >> 3);
-            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
-            const ushort_t* gaddr =
-                src + (long)(base0 + row) * ld + k0 + swz8;
-            ushort_t* laddr = lds + chunk * 512;
-            __builtin_amdgcn_global_load_lds(
-                (const __attribute__((address_space(1))) unsigned int*)gaddr,
-                (__attribute__((address_space(3))) unsigned int*)laddr,
-                16, 0, 0);
-        }
-    };
-
-    // prologue: stage up to 7 halves ahead
-    int next_half = 0;
-    while (next_half < 7 && next_half < total_halves) stage_half(next_half++);
-    if (nk >= 2)
-        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    else
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-
-    const int a_half = wm >> 7;            // this wave's fixed A half
-    const int b_half = wn >> 7;            // and B half
-    const int arow0 = wm & 127;
-    const int brow0 = wn & 127;
-
-    for (int kt = 0; kt < nk; kt++) {
-        const ushort_t* Ab = smem + (((kt & 1) << 2) | a_half) * (128 * FBK);
-        const ushort_t* Bb =
-            smem + (((kt & 1) << 2) | (2 | b_half)) * (128 * FBK);
-#pragma unroll
-        for (int q = 0; q < 4; q++) {      // quadrant: (k-half s, A-row half)
-            int sfr = q >> 1, ih = q & 1;
-            short8 af[4], bfr[4];
-#pragma unroll
-            for (int i = 0; i < 4; i++) {
-                int row = arow0 + ih * 64 + i * 16 + lrow;
-                int off = (sfr * 32 + kq * 8) ^ (swz_row(row) << 3);
-                af[i] = *(const short8*)&Ab[row * FBK + off];
-            }
-#pragma unroll
-            for (int j = 0; j < 4; j++) {
-                int row = brow0 + j * 16 + lrow;
-                int off = (sfr * 32 + kq * 8) ^ (swz_row(row) << 3);
-                bfr[j] = *(const short8*)&Bb[row * FBK + off];
-            }
-            if (next_half < total_halves) stage_half(next_half++);
-            __builtin_amdgcn_s_barrier();
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            __builtin_amdgcn_sched_barrier(0);
-            __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-            for (int i = 0; i < 4; i++)
-#pragma unroll
-                for (int j = 0; j < 4; j++)
-                    acc[ih * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af[i], bfr[j], acc[ih * 4 + i][j], 0, 0, 0);
-            __builtin_amdgcn_s_setprio(0);
-            __builtin_amdgcn_s_barrier();
-        }
-        // tile boundary: next tile's halves must have landed; keep at most
-        // the following tile's 3 halves (6 glds) in flight
-        if (kt + 1 < nk) {
-            if (kt + 2 < nk)
-                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __builtin_amdgcn_s_barrier();
-        }
-    }
-    __builtin_amdgcn_s_barrier();
-
-    // ---- epilogue: two half-tiles (rows 0-127, 128-255) via LDS ----
-    ushort_t* Cp;
-    long ldc;
-    {
-        const ushort_t* tmp;
-        OpArg ca;
-        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
-        ca.flags = p.Cflags;
-        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
-                     &tmp, &ldc);
-        Cp = (ushort_t*)tmp;
-    }
-    const ushort_t* biasp = nullptr;
-    if (p.has_bias)
-        biasp = (const ushort_t*)p.bias_base
-                + (long)(pid % p.nInner) * p.bias_sin
-                + (long)(pid / p.nInner) * p.bias_sout;
-    const float* csp = nullptr;
-    if (p.has_colscale)
-        csp = (const float*)p.colscale_base
-              + (long)(pid % p.nInner) * p.cs_sin
-              + (long)(pid / p.nInner) * p.cs_sout;
-    const ushort_t* auxp = nullptr;
-    if (p.epilogue == EPI_GELUGRAD)
-        auxp = (const ushort_t*)p.aux_base
-               + (long)(pid % p.nInner) * p.aux_sin
-               + (long)(pid / p.nInner) * p.aux_sout;
-    ushort_t* out2p = nullptr;
-    if (p.epilogue == EPI_GELU_PAIR)
-        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
-                + (long)(pid / p.nInner) * p.out2_sout;
-
-#pragma unroll
-    for (int half = 0; half < 2; half++) {
-        if (wm == half * 128) {
-            float csv[4] = {1.f, 1.f, 1.f, 1.f};
-            float bvv[4] = {0.f, 0.f, 0.f, 0.f};
-            if (csp) {
-#pragma unroll
-                for (int j16 = 0; j16 < 4; j16++)
-                    csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
-            }
-            if (biasp) {
-#pragma unroll
-                for (int j16 = 0; j16 < 4; j16++)
-                    bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
-            }
-#pragma unroll
-            for (int i16 = 0; i16 < 8; i16++) {
-#pragma unroll
-                for (int r = 0; r < 4; r++) {
-                    int li = i16 * 16 + kq * 4 + r;        // 0..127
-                    long gi = m0 + half * 128 + li;
-                    float vv[4];
-#pragma unroll
-                    for (int j16 = 0; j16 < 4; j16++)
-                        vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
-                    if (auxp) {
-                        const ushort_t* auxrow = auxp + gi * p.aux_ld;
-                        ushort_t av[4];
-#pragma unroll
-                        for (int j16 = 0; j16 < 4; j16++)
-                            av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
-#pragma unroll
-                        for (int j16 = 0; j16 < 4; j16++)
-                            vv[j16] *= gelu_grad_f(bf2f(av[j16]));
-                    }
-#pragma unroll
-                    for (int j16 = 0; j16 < 4; j16++) {
-                        int lj = wn + j16 * 16 + lrow;     // 0..255
-                        smem[li * EPI2_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
-                    }
-                }
-            }
-        }
-        __syncthreads();
-        {
-            int t = threadIdx.x;          // 512 threads: 128 rows x 4 qtrs
-            int li = t >> 2;
-            int qt = (t & 3) * 64;
-            long gi = m0 + half * 128 + li;
-            ushort_t* crow = Cp + gi * ldc + n0 + qt;
-            ushort_t* orow =
-                out2p ? out2p + gi * p.out2_ld + n0 + qt : nullptr;
-            const ushort_t* srow = smem + li * EPI2_ROW + qt;
-#pragma unroll
-            for (int c = 0; c < 8; c++) {
-                union { uint4v v; ushort_t u[8]; } x;
-                x.v = *(const uint4v*)(srow + c * 8);
-                *(uint4v*)(crow + c * 8) = x.v;
-                if (orow) {
-                    union { uint4v v; ushort_t u[8]; } g;
-#pragma unroll
-                    for (int e = 0; e < 8; e++)
-                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
-                    *(uint4v*)(orow + c * 8) = g.v;
-                }
-            }
-        }
-        __syncthreads();
-    }
-}
-
-void launch_gemm_nt_fast2(const GemmParams& p, hipStream_t stream) {
-    dim3 grid(p.N / BN2, p.M / BM2, p.nproblems);
-    hipLaunchKernelGGL(gemm_nt_fast2_kernel, grid, dim3(NT2), 0, stream, p);
-}
 
 
 // ---------------------------------------------------------------- //
@@ -638,6 +381,7 @@ void launch_gemm_nt_fast2(const GemmParams& p, hipStream_t stream) {
 
 #define NT3 512
 #define BN3 256
+#define EPI2_ROW 264   // 256 cols + 8 pad (16B-aligned LDS epilogue rows)
 
 __global__ __launch_bounds__(NT3) void gemm_nt_fast3_kernel(GemmParams p) {
     __shared__ ushort_t smem[2 * (128 + 256) * FBK];   // 96 KiB
@@ -977,166 +721,7 @@ void launch_gemm_nt_fast3(const GemmParams& p, hipStream_t stream) {
     hipLaunchKernelGGL(gemm_nt_fast3_kernel, grid, dim3(NT3), 0, stream, p);
 }
 
-// ---------------------------------------------------------------- //
-// 128x256-tile TN kernel (8 waves): both operands repack-staged at BK=64,
-// split-K capable. 2x the per-block work of tn_fast for the weight-grad
-// GEMMs; single 68 KiB LDS arena (2 blocks/CU).
 
-__global__ __launch_bounds__(NT3) void gemm_tn_fast2_kernel(GemmParams p) {
-    __shared__ ushort_t smem[128 * EPI2_ROW];    // >= A(128x64)+B(256x64)
-    ushort_t* As = smem;                          // [128][64]
-    ushort_t* Bs = smem + 128 * FBK;              // [256][64]
-
-    int pid = blockIdx.z;
-    int slice = 0, k_begin = 0, k_end = p.K;
-    if (p.splitk > 1) {
-        pid = blockIdx.z % p.nproblems;
-        slice = blockIdx.z / p.nproblems;
-        int per = ((p.K + FBK - 1) / FBK + p.splitk - 1) / p.splitk * FBK;
-        k_begin = slice * per;
-        k_end = min(p.K, k_begin + per);
-    }
-    int nwg = gridDim.x * gridDim.y;
-    int bid = blockIdx.y * gridDim.x + blockIdx.x;
-    {
-        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
-        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
-    }
-    const int n0 = (bid / gridDim.y) * BN3;
-    const int m0 = (bid % gridDim.y) * BM;
-
-    const ushort_t* Ap;
-    const ushort_t* Bp;
-    long lda, ldb;
-    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
-    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
-
-    const int wid = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int wm = (wid >> 2) * 64;
-    const int wn = (wid & 3) * 64;
-    const int lrow = lane & 15;
-    const int kq = lane >> 4;
-
-    f32x4 acc[4][4] = {};
-
-    for (int k0 = k_begin; k0 < k_end; k0 += FBK) {
-        // A: 128 cols -> 16x8=128 slots; B: 256 cols -> 32x8=256 slots
-        if (threadIdx.x < 128)
-            stage_repack(As, Ap, lda, k0, m0, k_end, threadIdx.x);
-        else if (threadIdx.x < 384) {
-            // 256 threads stage B: thread t covers col-block cb = t&31,
-            // m-block mb = t>>5 (same 8x8 micro-tile as stage_repack)
-            int t = threadIdx.x - 128;
-            int cb = t & 31;
-            int mb = t >> 5;
-            union { uint4v v; ushort_t u[8]; } rowv[8];
-#pragma unroll
-            for (int r = 0; r < 8; r++) {
-                int m = k0 + mb * 8 + r;
-                if (m < k_end)
-                    rowv[r].v = *(const uint4v*)(Bp + (long)m * ldb + n0
-                                                 + cb * 8);
-                else
-                    rowv[r].v = 0;
-            }
-#pragma unroll
-            for (int e = 0; e < 8; e++) {
-                union { uint4v v; ushort_t u[8]; } col;
-#pragma unroll
-                for (int r = 0; r < 8; r++) col.u[r] = rowv[r].u[e];
-                int row = cb * 8 + e;
-                int off = (mb * 8) ^ (swz_row(row) << 3);
-                *(uint4v*)&Bs[row * FBK + off] = col.v;
-            }
-        }
-        __syncthreads();
-        short8 af[2][4], bfr[2][4];
-#pragma unroll
-        for (int s = 0; s < 2; s++) {
-#pragma unroll
-            for (int i = 0; i < 4; i++) {
-                int row = wm + i * 16 + lrow;
-                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                af[s][i] = *(const short8*)&As[row * FBK + off];
-            }
-#pragma unroll
-            for (int j = 0; j < 4; j++) {
-                int row = wn + j * 16 + lrow;
-                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                bfr[s][j] = *(const short8*)&Bs[row * FBK + off];
-            }
-        }
-#pragma unroll
-        for (int s = 0; s < 2; s++)
-#pragma unroll
-            for (int i = 0; i < 4; i++)
-#pragma unroll
-                for (int j = 0; j < 4; j++)
-                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
-        __syncthreads();
-    }
-
-    if (p.splitk > 1) {
-        float* ws = p.ws + ((long)slice * p.nproblems + pid) * p.M * p.N;
-#pragma unroll
-        for (int i16 = 0; i16 < 4; i16++)
-#pragma unroll
-            for (int j16 = 0; j16 < 4; j16++) {
-                int j = n0 + wn + j16 * 16 + lrow;
-#pragma unroll
-                for (int r = 0; r < 4; r++) {
-                    int i = m0 + wm + i16 * 16 + kq * 4 + r;
-                    ws[(long)i * p.N + j] = acc[i16][j16][r];
-                }
-            }
-        return;
-    }
-    // plain epilogue via LDS staging (EPI_NONE only for TN)
-    ushort_t* Cp;
-    long ldc;
-    {
-        const ushort_t* tmp;
-        OpArg ca;
-        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
-        ca.flags = p.Cflags;
-        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
-                     &tmp, &ldc);
-        Cp = (ushort_t*)tmp;
-    }
-    __syncthreads();
-#pragma unroll
-    for (int i16 = 0; i16 < 4; i16++)
-#pragma unroll
-        for (int r = 0; r < 4; r++) {
-            int li = wm + i16 * 16 + kq * 4 + r;
-#pragma unroll
-            for (int j16 = 0; j16 < 4; j16++) {
-                int lj = wn + j16 * 16 + lrow;
-                smem[li * EPI2_ROW + lj] =
-                    f2bf(acc[i16][j16][r] * p.alpha);
-            }
-        }
-    __syncthreads();
-    {
-        int t = threadIdx.x;
-        int li = t >> 2;
-        int qt = (t & 3) * 64;
-        long gi = m0 + li;
-        ushort_t* crow = Cp + gi * ldc + n0 + qt;
-        const ushort_t* srow = smem + li * EPI2_ROW + qt;
-#pragma unroll
-        for (int c = 0; c < 8; c++)
-            *(uint4v*)(crow + c * 8) = *(const uint4v*)(srow + c * 8);
-    }
-}
-
-void launch_gemm_tn_fast2(const GemmParams& p, hipStream_t stream) {
-    int sk = p.splitk > 1 ? p.splitk : 1;
-    dim3 grid(p.N / BN3, p.M / BM, p.nproblems * sk);
-    hipLaunchKernelGGL(gemm_tn_fast2_kernel, grid, dim3(NT3), 0, stream, p);
-}
 
 // ------- 3-ring variant of nt_fast3 (counted vmcnt, raw barriers) ------
 __global__ __launch_bounds__(NT3) void gemm_nt_fast4_kernel(GemmParams p) {
@@ -1586,106 +1171,7 @@ void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream) {
     hipLaunchKernelGGL(gemm_nt_fast4_kernel, grid, dim3(NT3), 0, stream, p);
 }
 
-// ------- 3-ring variant of nt_fast (128^2, 1 block/CU) -------
-__global__ __launch_bounds__(NTHREADS) void gemm_nt_fast5_kernel(GemmParams p) {
-    // ONE shared array: a second __shared__ object makes hipcc emit
-    // s_waitcnt vmcnt(0) before the first ds_read of every k-step of a
-    // glds pipeline, draining the prefetch (guide §5 ".s-level traps" (a)).
-    __shared__ ushort_t smem[6 * BM * FBK];   // 3-ring x (A|B)
-    ushort_t* As0 = smem;
-    ushort_t* Bs0 = smem + 3 * BM * FBK;
 
-    const int pid = blockIdx.z;
-    // XCD-aware block remap (T1), column-major: each XCD die owns a
-    // contiguous run of N-columns (all M-tiles of a few n-tiles), so the
-    // shared B panel of a column stays resident in that XCD's private L2
-    // while the M sweep streams A.
-    int nwg = gridDim.x * gridDim.y;
-    int bid = blockIdx.y * gridDim.x + blockIdx.x;
-    {
-        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
-        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
-    }
-    const int n0 = (bid / gridDim.y) * BN;   // column-major: n outer
-    const int m0 = (bid % gridDim.y) * BM;
-
-    const ushort_t* Ap;
-    const ushort_t* Bp;
-    long lda, ldb;
-    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
-    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
-
-    const int wid = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int wm = (wid >> 1) * 64;
-    const int wn = (wid & 1) * 64;
-    const int lrow = lane & 15;
-    const int kq = lane >> 4;
-
-    f32x4 acc[4][4] = {};
-
-    const int nk = p.K / FBK;
-    stage_glds(As0, Ap, lda, m0, 0, wid, lane);
-    stage_glds(Bs0, Bp, ldb, n0, 0, wid, lane);
-    if (nk > 1) {
-        stage_glds(As0 + BM * FBK, Ap, lda, m0, FBK, wid, lane);
-        stage_glds(Bs0 + BM * FBK, Bp, ldb, n0, FBK, wid, lane);
-        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-    __builtin_amdgcn_s_barrier();
-
-    for (int kt = 0; kt < nk; kt++) {
-        int cur = kt % 3;
-        if (kt + 2 < nk) {
-            stage_glds(As0 + ((kt + 2) % 3) * BM * FBK, Ap, lda, m0,
-                       (kt + 2) * FBK, wid, lane);
-            stage_glds(Bs0 + ((kt + 2) % 3) * BM * FBK, Bp, ldb, n0,
-                       (kt + 2) * FBK, wid, lane);
-        }
-        short8 af[2][4], bfr[2][4];
-#pragma unroll
-        for (int s = 0; s < 2; s++) {
-#pragma unroll
-            for (int i = 0; i < 4; i++) {
-                int row = wm + i * 16 + lrow;
-                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                af[s][i] = *(const short8*)&As0[cur * BM * FBK + row * FBK + off];
-            }
-#pragma unroll
-            for (int j = 0; j < 4; j++) {
-                int row = wn + j * 16 + lrow;
-                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
-                bfr[s][j] = *(const short8*)&Bs0[cur * BN * FBK + row * FBK + off];
-            }
-        }
-#pragma unroll
-        for (int s = 0; s < 2; s++)
-#pragma unroll
-            for (int i = 0; i < 4; i++)
-#pragma unroll
-                for (int j = 0; j < 4; j++)
-                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
-        if (kt + 1 < nk) {
-            if (kt + 2 < nk)
-                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __builtin_amdgcn_s_barrier();
-        }
-    }
-    __syncthreads();
-    gemm_epilogue_lds(p, pid, m0, n0, wm, wn, lrow, kq, acc, smem);
-}
-
-
-void launch_gemm_nt_fast5(const GemmParams& p, hipStream_t stream) {
-    dim3 grid((p.N + BN - 1) / BN, (p.M + BM - 1) / BM, p.nproblems);
-    hipLaunchKernelGGL(gemm_nt_fast5_kernel, grid, dim3(NTHREADS), 0,
-                       stream, p);
-}
 
 // ------- split-K-only TN variant (32 KiB arena, 4 blocks/CU) -------
 __global__ __launch_bounds__(NTHREADS) void gemm_tn_sk_kernel(GemmParams p) {
@@ -1790,231 +1276,12 @@ void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
     hipLaunchKernelGGL(gemm_tn_sk_kernel, grid, dim3(NTHREADS), 0, stream, p);
 }
 
-
 // ---------------------------------------------------------------- //
-// 256x256-tile NT kernel, BK=32, 3-deep counted-vmcnt ring: the only
-// square (intensity-128 flops/byte) tile whose triple buffer fits LDS
-// (3 x 32 KiB). K-rows are stored PAIRED (two 64B k-rows per 128B LDS
-// row, linear for glds) so ds_read_b128 stays in the 64-bank regime;
-// the granule swizzle spreads the 16-lane read groups conflict-free.
-
-#define BK32 32
-
-__global__ __launch_bounds__(NT2) void gemm_nt_fast6_kernel(GemmParams p) {
-    __shared__ ushort_t smem[3 * 2 * BM2 * BK32];   // 96 KiB
-    const int tsz = BM2 * BK32;                     // elems per operand tile
-
-    const int pid = blockIdx.z;
-    int nwg = gridDim.x * gridDim.y;
-    int bid = blockIdx.y * gridDim.x + blockIdx.x;
-    {
-        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
-        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
-    }
-    const int n0 = (bid / gridDim.y) * BN2;
-    const int m0 = (bid % gridDim.y) * BM2;
-
-    const ushort_t* Ap;
-    const ushort_t* Bp;
-    long lda, ldb;
-    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
-    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
-
-    const int wid = threadIdx.x / WAVE;
-    const int lane = threadIdx.x % WAVE;
-    const int wm = (wid >> 2) * 128;
-    const int wn = (wid & 3) * 64;
-    const int lrow = lane & 15;
-    const int kq = lane >> 4;
-
-    f32x4 acc[8][4] = {};
-
-    // stage one K-step (256 rows x 32 k per operand = 16 chunks, 2 per
-    // wave per operand). Paired-row image: k-rows 2R,2R+1 share LDS row R.
-    auto stage = [&](int buf, int k0) {
-#pragma unroll
-        for (int op = 0; op < 2; op++) {
-            const ushort_t* src = op ? Bp : Ap;
-            long ld = op ? ldb : lda;
-            int base0 = op ? n0 : m0;
-            ushort_t* lds = smem + (buf * 2 + op) * tsz;
-#pragma unroll
-            for (int c = 0; c < 2; c++) {
-                int chunk = wid * 2 + c;
-                int R = chunk * 8 + (lane >> 3);        // image row (128B)
-                int gg = (lane & 7) ^ swz_row(R);       // swizzled granule
-                int row = R * 2 + (gg >> 2);            // global tile row
-                int kpart = (gg & 3) * 8;
-                const ushort_t* gaddr =
-                    src + (long)(base0 + row) * ld + k0 + kpart;
-                ushort_t* laddr = lds + chunk * 512;
-                __builtin_amdgcn_global_load_lds(
-                    (const __attribute__((address_space(1))) unsigned int*)
-                        gaddr,
-                    (__attribute__((address_space(3))) unsigned int*)laddr,
-                    16, 0, 0);
-            }
-        }
-    };
-
-    const int nk = p.K / BK32;
-    stage(0, 0);
-    if (nk > 1) stage(1, BK32);
-    if (nk > 1)
-        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    else
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-
-    for (int kt = 0; kt < nk; kt++) {
-        if (kt + 2 < nk) stage((kt + 2) % 3, (kt + 2) * BK32);
-        const ushort_t* Al = smem + ((kt % 3) * 2 + 0) * tsz;
-        const ushort_t* Bl = smem + ((kt % 3) * 2 + 1) * tsz;
-        short8 bfr[4];
-#pragma unroll
-        for (int j = 0; j < 4; j++) {
-            int row = wn + j * 16 + lrow;
-            int R = row >> 1;
-            int gg = ((row & 1) * 4 + kq) ^ swz_row(R);
-            bfr[j] = *(const short8*)&Bl[R * 64 + gg * 8];
-        }
-#pragma unroll
-        for (int i = 0; i < 8; i++) {
-            int row = wm + i * 16 + lrow;
-            int R = row >> 1;
-            int gg = ((row & 1) * 4 + kq) ^ swz_row(R);
-            short8 af = *(const short8*)&Al[R * 64 + gg * 8];
-#pragma unroll
-            for (int j = 0; j < 4; j++)
-                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    af, bfr[j], acc[i][j], 0, 0, 0);
-        }
-        if (kt + 1 < nk) {
-            if (kt + 2 < nk)
-                asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-            else
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __builtin_amdgcn_s_barrier();
-        }
-    }
-    __syncthreads();
-
-    // ---- epilogue: two half-tiles (rows 0-127, 128-255) via LDS ----
-    ushort_t* Cp;
-    long ldc;
-    {
-        const ushort_t* tmp;
-        OpArg ca;
-        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
-        ca.flags = p.Cflags;
-        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
-                     &tmp, &ldc);
-        Cp = (ushort_t*)tmp;
-    }
-    const ushort_t* biasp = nullptr;
-    if (p.has_bias)
-        biasp = (const ushort_t*)p.bias_base
-                + (long)(pid % p.nInner) * p.bias_sin
-                + (long)(pid / p.nInner) * p.bias_sout;
-    const float* csp = nullptr;
-    if (p.has_colscale)
-        csp = (const float*)p.colscale_base
-              + (long)(pid % p.nInner) * p.cs_sin
-              + (long)(pid / p.nInner) * p.cs_sout;
-    const ushort_t* auxp = nullptr;
-    if (p.epilogue == EPI_GELUGRAD)
-        auxp = (const ushort_t*)p.aux_base
-               + (long)(pid % p.nInner) * p.aux_sin
-               + (long)(pid / p.nInner) * p.aux_sout;
-    ushort_t* out2p = nullptr;
-    if (p.epilogue == EPI_GELU_PAIR)
-        out2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
-                + (long)(pid / p.nInner) * p.out2_sout;
-
-#pragma unroll
-    for (int half = 0; half < 2; half++) {
-        if (wm == half * 128) {
-            float csv[4] = {1.f, 1.f, 1.f, 1.f};
-            float bvv[4] = {0.f, 0.f, 0.f, 0.f};
-            if (csp) {
-#pragma unroll
-                for (int j16 = 0; j16 < 4; j16++)
-                    csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
-            }
-            if (biasp) {
-#pragma unroll
-                for (int j16 = 0; j16 < 4; j16++)
-                    bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
-            }
-#pragma unroll
-            for (int i16 = 0; i16 < 8; i16++) {
-#pragma unroll
-                for (int r = 0; r < 4; r++) {
-                    int li = i16 * 16 + kq * 4 + r;        // 0..127
-                    long gi = m0 + half * 128 + li;
-                    float vv[4];
-#pragma unroll
-                    for (int j16 = 0; j16 < 4; j16++)
-                        vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
-                    if (auxp) {
-                        const ushort_t* auxrow = auxp + gi * p.aux_ld;
-                        ushort_t av[4];
-#pragma unroll
-                        for (int j16 = 0; j16 < 4; j16++)
-                            av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
-#pragma unroll
-                        for (int j16 = 0; j16 < 4; j16++)
-                            vv[j16] *= gelu_grad_f(bf2f(av[j16]));
-                    }
-#pragma unroll
-                    for (int j16 = 0; j16 < 4; j16++) {
-                        int lj = wn + j16 * 16 + lrow;     // 0..255
-                        smem[li * EPI2_ROW + lj] = f2bf(vv[j16] + bvv[j16]);
-                    }
-                }
-            }
-        }
-        __syncthreads();
-        {
-            int t = threadIdx.x;          // 512 threads: 128 rows x 4 qtrs
-            int li = t >> 2;
-            int qt = (t & 3) * 64;
-            long gi = m0 + half * 128 + li;
-            ushort_t* crow = Cp + gi * ldc + n0 + qt;
-            ushort_t* orow =
-                out2p ? out2p + gi * p.out2_ld + n0 + qt : nullptr;
-            const ushort_t* srow = smem + li * EPI2_ROW + qt;
-#pragma unroll
-            for (int c = 0; c < 8; c++) {
-                union { uint4v v; ushort_t u[8]; } x;
-                x.v = *(const uint4v*)(srow + c * 8);
-                *(uint4v*)(crow + c * 8) = x.v;
-                if (orow) {
-                    union { uint4v v; ushort_t u[8]; } g;
-#pragma unroll
-                    for (int e = 0; e < 8; e++)
-                        g.u[e] = f2bf(gelu_f(bf2f(x.u[e])));
-                    *(uint4v*)(orow + c * 8) = g.v;
-                }
-            }
-        }
-        if (p.colsum_out) {
-            float* outp = p.colsum_out
-                + (long)(pid % p.nInner) * p.colsum_sin;
-            int c = threadIdx.x;
-            if (c < BN2) {
-                float ssum = 0.f;
-                for (int r = 0; r < 128; r++)
-                    ssum += bf2f(smem[r * EPI2_ROW + c]);
-                atomicAdd(&outp[n0 + c], ssum);
-            }
-        }
-        __syncthreads();
-    }
-}
-
-
-void launch_gemm_nt_fast6(const GemmParams& p, hipStream_t stream) {
-    dim3 grid(p.N / BN2, p.M / BM2, p.nproblems);
-    hipLaunchKernelGGL(gemm_nt_fast6_kernel, grid, dim3(NT2), 0, stream, p);
-}
+// Measured-and-rejected variants (see git history + profiles/README.md):
+//  - 256^2 2-phase and phase-pipelined kernels (1 block/CU sync stalls
+//    beat the intensity gain: 670-774us vs 463us for the up shape)
+//  - 128^2 3-ring (96 KiB LDS halves occupancy: down 292->357us)
+//  - 128x256 TN (68 KiB arena halves tn_fast's 4 blocks/CU: dW1 +37%)
+//  - 256^2 BK=32 3-ring (paired-row LDS; slower on K=512 and buggy)
+// The shipping set: gemm_nt_fast (128^2 glds 2-buf), gemm_nt_fast3/4
+// (128x256 8-wave, 2-buf / 3-ring), gemm_tn_fast(+_sk), gemm_nn_fast.
