@@ -285,3 +285,27 @@ def test_denoising_trainer_gpu_step():
     assert l1 > 0 and l2 > 0
     for n, p in m.named_parameters():
         assert torch.isfinite(p.float()).all(), n
+
+
+def test_levels2_minimum_native():
+    """levels=2: top_down has a single group (G=1) — smallest table case."""
+    torch.manual_seed(0)
+    m32 = Glom(dim=64, levels=2, image_size=32, patch_size=8).to(DEV)
+    m32.force_eager = True
+    mbf = Glom(dim=64, levels=2, image_size=32, patch_size=8).to(DEV)
+    mbf.load_state_dict(m32.state_dict())
+    mbf = mbf.to(torch.bfloat16)
+    img = torch.randn(2, 3, 32, 32, device=DEV)
+    ref = m32(img, iters=3)
+    out = mbf(img.to(torch.bfloat16), iters=3)
+    assert _rel_err(out, ref) < 2e-2
+
+
+def test_batch1_native():
+    torch.manual_seed(0)
+    m = Glom(dim=512, levels=6, image_size=224, patch_size=14)
+    m = m.to(DEV, torch.bfloat16)
+    img = torch.randn(1, 3, 224, 224, device=DEV, dtype=torch.bfloat16)
+    out = m(img, iters=12)
+    assert out.shape == (1, 256, 6, 512)
+    assert torch.isfinite(out.float()).all()
