@@ -109,3 +109,26 @@ def test_gradients_flow():
     assert m.top_down.net[3].weight.grad is not None
     assert m.pos_emb.weight.grad is not None
     assert m.image_to_tokens[1].weight.grad is not None
+
+
+def test_fuzz_configs_vs_oracle():
+    """Randomized config sweep vs the loop-level oracle (shapes, flags)."""
+    import random
+    rng = random.Random(7)
+    for trial in range(6):
+        levels = rng.choice([2, 3, 4])
+        dim = rng.choice([16, 24, 40])
+        patch = rng.choice([4, 8])
+        side = rng.choice([2, 3])
+        size = patch * side
+        kw = dict(dim=dim, levels=levels, image_size=size, patch_size=patch,
+                  consensus_self=rng.random() < 0.5)
+        if rng.random() < 0.3:
+            kw["local_consensus_radius"] = 1
+        torch.manual_seed(trial)
+        m = Glom(**kw)
+        img = torch.randn(rng.choice([1, 2]), 3, size, size)
+        it = rng.choice([1, 2, 3])
+        out = m(img, iters=it)
+        ref = oracle_forward(m, img, iters=it)
+        assert torch.allclose(out, ref, rtol=5e-5, atol=5e-5), (kw, it)
