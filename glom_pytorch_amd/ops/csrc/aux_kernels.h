@@ -21,3 +21,4 @@ void launch_mix_bwd(const void* dout, void* dmix, void* dtd, long total,
                     int L, int d, hipStream_t s);
 void launch_add4(const void* a, const void* b, const void* c, const void* d,
                  void* out, long total, hipStream_t s);
+void launch_gelu(const void* in, void* out, long total, hipStream_t s);

@@ -423,3 +423,28 @@ void launch_add4(const void* a, const void* b, const void* c, const void* d,
                        (const ushort_t*)b, (const ushort_t*)c,
                        (const ushort_t*)d, (ushort_t*)out, total);
 }
+
+// out = gelu(in), elementwise bf16 (the up-projection's activation pass;
+// running this as its own bandwidth-bound kernel beats computing it in the
+// GEMM epilogue, which runs at 1 block/CU with nothing to overlap)
+__global__ __launch_bounds__(NTHREADS) void k_gelu(
+        const ushort_t* __restrict__ in, ushort_t* __restrict__ out,
+        long total) {
+    long i8 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
+    if (i8 >= total) return;
+    union { uint4v v; ushort_t u[8]; } x, o;
+    x.v = *(const uint4v*)(in + i8);
+    float xin[8], yv[8];
+#pragma unroll
+    for (int e = 0; e < 8; e++) xin[e] = bf2f(x.u[e]);
+    gelu_f_vec<8>(xin, yv);
+#pragma unroll
+    for (int e = 0; e < 8; e++) o.u[e] = f2bf(yv[e]);
+    *(uint4v*)(out + i8) = o.v;
+}
+
+void launch_gelu(const void* in, void* out, long total, hipStream_t s) {
+    hipLaunchKernelGGL(k_gelu, dim3(cdiv(total / 8, NTHREADS)),
+                       dim3(NTHREADS), 0, s, (const ushort_t*)in,
+                       (ushort_t*)out, total);
+}

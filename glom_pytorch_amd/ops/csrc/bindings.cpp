@@ -156,9 +156,11 @@ std::vector<torch::Tensor> grouped_ff_fwd(
         p.B.base = w1.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
         p.Cbase = Hpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
         p.bias_base = b1.data_ptr(); p.bias_sin = m4; p.has_bias = 1;
-        p.epilogue = EPI_GELU_PAIR;
-        p.out2 = Hact.data_ptr(); p.out2_sin = M * m4; p.out2_ld = m4;
         run_gemm(p, s, opts, true);
+        // activation as its own bandwidth-bound pass: measured faster than
+        // the GEMM-epilogue fusion (which runs at 1 block/CU)
+        launch_gelu(Hpre.data_ptr(), Hact.data_ptr(), Hpre.numel(), s);
+        check_launch();
     }
     // down-projection: Y_g = Hact_g @ W2_g^T + b2_g
     {
