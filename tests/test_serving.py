@@ -1,0 +1,54 @@
+"""Serving layer: micro-batched embed endpoint (CPU eager model here; the
+same app serves the CDNA4 engine under hipGraph replay on GPU)."""
+
+import numpy as np
+import pytest
+import torch
+
+from glom_pytorch_amd import Glom
+from conftest import SMALL
+
+fastapi = pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient  # noqa: E402
+
+from glom_pytorch_amd.serving import create_app  # noqa: E402
+
+
+def test_embed_endpoint_matches_direct_forward():
+    torch.manual_seed(0)
+    model = Glom(**SMALL)
+    app = create_app(model, iters=3, max_wait_ms=1.0)
+    with TestClient(app) as client:
+        r = client.get("/healthz")
+        assert r.status_code == 200 and r.json()["ok"]
+
+        img = np.random.RandomState(0).randn(3, 32, 32).astype(np.float32)
+        r = client.post("/embed", content=img.tobytes())
+        assert r.status_code == 200
+        n, d = map(int, r.headers["x-shape"].split(","))
+        out = np.frombuffer(r.content, dtype=np.float32).reshape(n, d)
+
+        with torch.no_grad():
+            ref = model(torch.from_numpy(img)[None], iters=3)[0, :, -1]
+        assert np.allclose(out, ref.numpy(), atol=1e-5)
+
+
+def test_embed_microbatching_concurrent():
+    import concurrent.futures
+    torch.manual_seed(0)
+    model = Glom(**SMALL)
+    app = create_app(model, iters=2, max_batch=4, max_wait_ms=20.0)
+    imgs = [np.random.RandomState(i).randn(3, 32, 32).astype(np.float32)
+            for i in range(6)]
+    with TestClient(app) as client:
+        def post(i):
+            return client.post("/embed", content=imgs[i].tobytes())
+        with concurrent.futures.ThreadPoolExecutor(6) as ex:
+            rs = list(ex.map(post, range(6)))
+        for i, r in enumerate(rs):
+            assert r.status_code == 200
+            n, d = map(int, r.headers["x-shape"].split(","))
+            out = np.frombuffer(r.content, dtype=np.float32).reshape(n, d)
+            with torch.no_grad():
+                ref = model(torch.from_numpy(imgs[i])[None], iters=2)[0, :, -1]
+            assert np.allclose(out, ref.numpy(), atol=1e-5), i
