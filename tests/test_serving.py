@@ -52,3 +52,21 @@ def test_embed_microbatching_concurrent():
             with torch.no_grad():
                 ref = model(torch.from_numpy(imgs[i])[None], iters=2)[0, :, -1]
             assert np.allclose(out, ref.numpy(), atol=1e-5), i
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_embed_endpoint_gpu_graphs():
+    torch.manual_seed(0)
+    model = Glom(**SMALL).to("cuda", torch.bfloat16).enable_graphs()
+    app = create_app(model, iters=3, max_wait_ms=1.0)
+    with TestClient(app) as client:
+        img = np.random.RandomState(1).randn(3, 32, 32).astype(np.float32)
+        r = client.post("/embed", content=img.tobytes())
+        assert r.status_code == 200
+        n, d = map(int, r.headers["x-shape"].split(","))
+        out = np.frombuffer(r.content, dtype=np.float32).reshape(n, d)
+        with torch.no_grad():
+            ref = model(torch.from_numpy(img)[None].to("cuda",
+                        torch.bfloat16), iters=3)[0, :, -1]
+        assert np.allclose(out, ref.float().cpu().numpy(), atol=1e-2)
