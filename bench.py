@@ -89,7 +89,9 @@ def main():
         trainer = DenoisingTrainer(model, distributed=distributed)
 
         def step(x, iters):
-            return trainer.step(x, iters=iters)
+            # loss stays on-device inside the timed loop (no host sync);
+            # converted once after the final synchronize
+            return trainer.step(x, iters=iters, sync_loss=False)
 
     for _ in range(args.warmup):
         step(img, args.iters)
@@ -106,6 +108,8 @@ def main():
             loss = step(img, args.iters)
     sync_all()
     elapsed = time.perf_counter() - t0
+    if torch.is_tensor(loss):
+        loss = loss.item()
 
     # max over ranks
     if distributed:

@@ -56,7 +56,11 @@ class DenoisingTrainer:
         self.step_idx = 0
         self.log_path = log_path
 
-    def step(self, img: torch.Tensor, iters: int | None = None) -> float:
+    def step(self, img: torch.Tensor, iters: int | None = None,
+             sync_loss: bool = True):
+        """One denoising training step. With sync_loss=False the loss is
+        returned as a device tensor (no host sync), letting the host run
+        ahead and queue the next step's launches."""
         iters = iters if iters is not None else 2 * self.model.levels
         t = min(self.decode_step, iters)
         self.opt.zero_grad(set_to_none=True)
@@ -72,7 +76,7 @@ class DenoisingTrainer:
             self.heartbeat.tick()
         self.opt.step()
         self.step_idx += 1
-        return loss.item()
+        return loss.item() if sync_loss else loss.detach()
 
     def log(self, **metrics):
         if self.log_path:
