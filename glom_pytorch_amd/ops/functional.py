@@ -107,30 +107,37 @@ class GlomStepFn(torch.autograd.Function):
                 tw1, tb1, tw2, tb2, attend_self, mask,
                 bw1t, bw2t, tw1t, tw2t):
         ext = _load_extension()
-        # the three per-iteration chains (bottom-up MLP, top-down MLP,
-        # consensus attention) only depend on `levels`: fork them onto
-        # side streams so their tail waves overlap, join before the mix
-        cur = torch.cuda.current_stream()
-        s_td, s_at = GlomStepFn._side_streams()
-        ev = torch.cuda.Event()
-        ev.record(cur)
-        buY, bhp, bha = ext.grouped_ff_fwd(tokens, levels, None,
-                                           bw1, bb1, bw2, bb2, 0)
-        with torch.cuda.stream(s_td):
-            s_td.wait_event(ev)
-            tdY, thp, tha = ext.grouped_ff_fwd(None, levels, pos,
-                                               tw1, tb1, tw2, tb2, 1)
-        with torch.cuda.stream(s_at):
-            s_at.wait_event(ev)
-            cons, probs, rnorm = ext.consensus_fwd(levels, attend_self,
-                                                   mask)
-        cur.wait_stream(s_td)
-        cur.wait_stream(s_at)
-        # boundary tensors crossing back to the main stream: pin their
-        # blocks until the main stream catches up (allocator safety)
-        for t in (tdY, thp, tha, cons, probs, rnorm):
-            t.record_stream(cur)
-        out = ext.level_mix_fwd(levels, buY, tdY, cons)
+        if torch.is_grad_enabled():
+            out, bhp, bha, thp, tha, probs, rnorm = ext.glom_step_fwd(
+                tokens, levels, pos, bw1, bb1, bw2, bb2, tw1, tb1, tw2,
+                tb2, attend_self, mask)
+        else:
+            # inference: the three per-iteration chains (bottom-up MLP,
+            # top-down MLP, consensus attention) only depend on `levels`;
+            # forking them onto side streams overlaps their tail waves
+            # (+10% forward throughput). Training keeps one stream: the
+            # backward-dominated step measured no gain from the fork.
+            cur = torch.cuda.current_stream()
+            s_td, s_at = GlomStepFn._side_streams()
+            ev = torch.cuda.Event()
+            ev.record(cur)
+            buY, bhp, bha = ext.grouped_ff_fwd(tokens, levels, None,
+                                               bw1, bb1, bw2, bb2, 0)
+            with torch.cuda.stream(s_td):
+                s_td.wait_event(ev)
+                tdY, thp, tha = ext.grouped_ff_fwd(None, levels, pos,
+                                                   tw1, tb1, tw2, tb2, 1)
+            with torch.cuda.stream(s_at):
+                s_at.wait_event(ev)
+                cons, probs, rnorm = ext.consensus_fwd(levels, attend_self,
+                                                       mask)
+            cur.wait_stream(s_td)
+            cur.wait_stream(s_at)
+            # boundary tensors crossing back to the main stream: pin their
+            # blocks until the main stream catches up (allocator safety)
+            for t in (tdY, thp, tha, cons, probs, rnorm):
+                t.record_stream(cur)
+            out = ext.level_mix_fwd(levels, buY, tdY, cons)
         ctx.save_for_backward(tokens, levels, pos, bw1, bw2, tw1, tw2,
                               bhp, bha, thp, tha, probs, rnorm, mask,
                               bw1t, bw2t, tw1t, tw2t)
