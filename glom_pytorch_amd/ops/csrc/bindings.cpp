@@ -610,6 +610,14 @@ double bench_gemm(int64_t M, int64_t N, int64_t K, int64_t layout,
     return ms / reps;
 }
 
+void add4_into(torch::Tensor a, torch::Tensor b, torch::Tensor c,
+               torch::Tensor d, torch::Tensor out) {
+    CHECK_IN(a); CHECK_IN(b); CHECK_IN(c); CHECK_IN(d); CHECK_IN(out);
+    launch_add4(a.data_ptr(), b.data_ptr(), c.data_ptr(), d.data_ptr(),
+                out.data_ptr(), out.numel(), cur_stream());
+    check_launch();
+}
+
 std::string build_info() {
     return "glom_pytorch_amd HIP extension (gfx950, bf16 MFMA 16x16x32)";
 }
@@ -630,6 +638,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("level_mix_bwd", &level_mix_bwd, "level mix backward");
     m.def("glom_step_fwd", &glom_step_fwd, "full GLOM iteration forward");
     m.def("glom_step_bwd", &glom_step_bwd, "full GLOM iteration backward");
+    m.def("add4_into", &add4_into, "fused 4-way elementwise sum");
     m.def("build_info", &build_info);
     m.def("bench_gemm", &bench_gemm, "raw GEMM microbench (tuning only)");
 }

@@ -9,6 +9,8 @@ reference's denoising recipe, README.md:56-90).
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from glom_pytorch_amd.ops import _load_extension
@@ -149,13 +151,43 @@ class GlomStepFn(torch.autograd.Function):
         ext = _load_extension()
         (tokens, levels, pos, bw1, bw2, tw1, tw2, bhp, bha, thp, tha,
          probs, rnorm, mask, bw1t, bw2t, tw1t, tw2t) = ctx.saved_tensors
-        (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
-         dtw1, dtb1, dtw2, dtb2) = ext.glom_step_bwd(
-            dnew.contiguous(), tokens, levels, pos, bw1, bw2, tw1, tw2,
-            bhp, bha, thp, tha, probs, rnorm, ctx.attend_self, mask,
-            bw1t, bw2t, tw1t, tw2t)
-        return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
-                dtw1, dtb1, dtw2, dtb2, None, None, None, None, None, None)
+        if os.environ.get("GLOM_BWD_STREAMS", "0") != "1":
+            (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
+             dtw1, dtb1, dtw2, dtb2) = ext.glom_step_bwd(
+                dnew.contiguous(), tokens, levels, pos, bw1, bw2, tw1, tw2,
+                bhp, bha, thp, tha, probs, rnorm, ctx.attend_self, mask,
+                bw1t, bw2t, tw1t, tw2t)
+            return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
+                    dtw1, dtb1, dtw2, dtb2, None, None, None, None, None,
+                    None)
+        # experimental: fork the three independent backward chains
+        cur = torch.cuda.current_stream()
+        s_td, s_at = GlomStepFn._side_streams()
+        dmix, dtd = ext.level_mix_bwd(dnew.contiguous())
+        ev = torch.cuda.Event()
+        ev.record(cur)
+        bu = ext.grouped_ff_bwd(dmix, tokens, levels, None, bw1, bw2,
+                                bhp, bha, 0, bw1t, bw2t)
+        with torch.cuda.stream(s_td):
+            s_td.wait_event(ev)
+            td = ext.grouped_ff_bwd(dtd, None, levels, pos, tw1, tw2,
+                                    thp, tha, 1, tw1t, tw2t)
+        with torch.cuda.stream(s_at):
+            s_at.wait_event(ev)
+            dAttn = ext.consensus_bwd(dmix, levels, probs, rnorm,
+                                      ctx.attend_self, mask)
+        cur.wait_stream(s_td)
+        cur.wait_stream(s_at)
+        for t in list(td) + [dAttn]:
+            if t is not None and t.numel():
+                t.record_stream(cur)
+        from glom_pytorch_amd.ops import _load_extension as _le
+        dLevels = torch.empty_like(levels)
+        _le().add4_into(dmix, bu[1], td[1], dAttn, dLevels)
+        dPos = td[1][:, :, 1:, :].sum(dim=(0, 2))
+        return (bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
+                td[2], td[3], td[4], td[5], None, None, None, None, None,
+                None)
 
 
 def _transposed_weights(model):
