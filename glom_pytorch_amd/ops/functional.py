@@ -109,7 +109,8 @@ class GlomStepFn(torch.autograd.Function):
                 tw1, tb1, tw2, tb2, attend_self, mask,
                 bw1t, bw2t, tw1t, tw2t):
         ext = _load_extension()
-        if torch.is_grad_enabled():
+        if (torch.is_grad_enabled()
+                and os.environ.get("GLOM_FWD_STREAMS", "0") != "1"):
             out, bhp, bha, thp, tha, probs, rnorm = ext.glom_step_fwd(
                 tokens, levels, pos, bw1, bb1, bw2, bb2, tw1, tb1, tw2,
                 tb2, attend_self, mask)
