@@ -309,3 +309,27 @@ def test_batch1_native():
     out = m(img, iters=12)
     assert out.shape == (1, 256, 6, 512)
     assert torch.isfinite(out.float()).all()
+
+
+def test_backward_singlestream_path_matches():
+    """The opt-out single-stream mega-backward must agree with the default
+    stream-forked backward."""
+    import os
+    torch.manual_seed(0)
+    m = Glom(**CFG).to(DEV, torch.bfloat16)
+    img = torch.randn(2, 3, 32, 32, device=DEV, dtype=torch.bfloat16)
+
+    def grads():
+        m.zero_grad()
+        out = m(img, iters=3, return_all=True)
+        out[-1].float().pow(2).mean().backward()
+        return {n: p.grad.clone() for n, p in m.named_parameters()}
+
+    g_fork = grads()
+    os.environ["GLOM_NO_BWD_STREAMS"] = "1"
+    try:
+        g_single = grads()
+    finally:
+        del os.environ["GLOM_NO_BWD_STREAMS"]
+    for n in g_fork:
+        assert torch.equal(g_fork[n], g_single[n]), n
