@@ -332,4 +332,10 @@ def test_backward_singlestream_path_matches():
     finally:
         del os.environ["GLOM_NO_BWD_STREAMS"]
     for n in g_fork:
-        assert torch.equal(g_fork[n], g_single[n]), n
+        if n.endswith("net.1.bias"):
+            # dB1 is an f32 atomic column sum: summation order (and hence
+            # the bf16 rounding) varies run to run — compare with tolerance
+            a, b = g_fork[n].float(), g_single[n].float()
+            assert ((a - b).norm() / b.norm().clamp_min(1e-8)) < 1e-2, n
+        else:
+            assert torch.equal(g_fork[n], g_single[n]), n
