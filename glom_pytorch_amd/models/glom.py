@@ -14,9 +14,10 @@ Compute dispatch (this file is the thin PyTorch-ROCm layer; all heavy lifting
 is below it):
   - CPU / fp32, or ``GLOM_FORCE_EAGER=1``: a plain-PyTorch eager path.
   - bf16 tensors on a gfx950 GPU: hand-written CDNA4 HIP kernels via
-    ``glom_pytorch_amd.ops`` (MFMA grouped GEMMs with fused GELU / pos-emb,
-    fused consensus attention, fused level mixing). This path raises if the
-    HIP extension is not importable on a GPU machine — there is no silent
+    ``glom_pytorch_amd.ops`` (grouped MFMA GEMM family, fully fused
+    consensus attention incl. masked softmax and AV, fused level mixing,
+    hand-written backward for everything). This path raises if the HIP
+    extension is not importable on a GPU machine — there is no silent
     eager fallback on GPU.
 """
 

@@ -21,8 +21,8 @@ class GroupedFFFn(torch.autograd.Function):
 
     mode 0 (bottom-up): group 0 consumes `tokens`, group g consumes
     levels[..., g-1, :]  (reference glom_pytorch.py:132-134, without the cat)
-    mode 1 (top-down): group g consumes levels[..., g+1, :] + pos, fused into
-    the GEMM A-operand load (reference glom_pytorch.py:136).
+    mode 1 (top-down): group g consumes levels[..., g+1, :] + pos; the pos
+    add is materialized once per call by k_add_pos (glom_pytorch.py:136).
     """
 
     @staticmethod
