@@ -618,6 +618,126 @@ void add4_into(torch::Tensor a, torch::Tensor b, torch::Tensor c,
     check_launch();
 }
 
+
+// ------------------------------------------------------------------ //
+// Fine-grained FF backward stages (stream-forked by the python layer):
+// dH (with fused dB1), then dX and dW independently.
+
+std::vector<torch::Tensor> ff_bwd_dh(torch::Tensor dY, torch::Tensor w2t,
+                                     torch::Tensor Hpre) {
+    CHECK_IN(dY); CHECK_IN(w2t); CHECK_IN(Hpre);
+    const int64_t G = Hpre.size(0), M = Hpre.size(1), m4 = Hpre.size(2);
+    const int64_t d = dY.size(3);
+    auto opts = dY.options();
+    hipStream_t s = cur_stream();
+    auto dHpre = torch::empty({G, M, m4}, opts);
+    const bool fused = (M % 128 == 0) && (m4 % 256 == 0) && (m4 >= 1024)
+                       && (d % 64 == 0);
+    torch::Tensor db1f;
+    GemmParams p = base_params(M, m4, d, LAYOUT_NT, G, G, 1.0f);
+    p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
+    p.B.base = w2t.data_ptr(); p.B.sin = m4 * d; p.B.ld = d;
+    p.Cbase = dHpre.data_ptr(); p.Csin = M * m4; p.Cld = m4;
+    p.epilogue = EPI_GELUGRAD;
+    p.aux_base = Hpre.data_ptr(); p.aux_sin = M * m4; p.aux_ld = m4;
+    if (fused) {
+        db1f = torch::zeros({G, m4}, opts.dtype(at::kFloat));
+        p.colsum_out = db1f.data_ptr<float>();
+        p.colsum_sin = m4;
+    }
+    run_gemm(p, s, opts, true);
+    auto dB1 = fused ? db1f.flatten().to(at::kBFloat16)
+                     : dHpre.sum(1).flatten();
+    return {dHpre, dB1};
+}
+
+std::vector<torch::Tensor> ff_bwd_dx(torch::Tensor dHpre, torch::Tensor w1t,
+                                     c10::optional<torch::Tensor> tokens_opt,
+                                     int64_t B, int64_t N, int64_t L,
+                                     int64_t mode) {
+    CHECK_IN(dHpre); CHECK_IN(w1t);
+    const int64_t G = dHpre.size(0), M = dHpre.size(1), m4 = dHpre.size(2);
+    const int64_t d = w1t.numel() / (G * m4);
+    auto opts = dHpre.options();
+    hipStream_t s = cur_stream();
+    torch::Tensor dTokens;
+    auto dLevels = torch::zeros({B, N, L, d}, opts);
+    GemmParams p = base_params(M, d, m4, LAYOUT_NT, G, G, 1.0f);
+    p.A.base = dHpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
+    p.B.base = w1t.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
+    if (mode == 0) {
+        auto tokens = tokens_opt.value();
+        CHECK_IN(tokens);
+        dTokens = torch::empty({B, N, d}, opts);
+        p.Cflags = OP_TABLE;
+        p.Ctab[0] = dTokens.data_ptr();
+        p.Ctabld[0] = d;
+        for (int64_t g = 1; g < G; g++) {
+            p.Ctab[g] = (char*)dLevels.data_ptr() + (g - 1) * d * 2;
+            p.Ctabld[g] = L * d;
+        }
+    } else {
+        p.Cbase = (char*)dLevels.data_ptr() + d * 2;
+        p.Csin = d; p.Cld = L * d;
+    }
+    run_gemm(p, s, opts, true);
+    if (mode != 0) dTokens = torch::empty({0}, opts);
+    return {dTokens, dLevels};
+}
+
+std::vector<torch::Tensor> ff_bwd_dw(
+        torch::Tensor dY, torch::Tensor dHpre,
+        c10::optional<torch::Tensor> tokens_opt, torch::Tensor levels,
+        c10::optional<torch::Tensor> pos_opt, torch::Tensor Hact,
+        int64_t mode) {
+    CHECK_IN(dY); CHECK_IN(dHpre); CHECK_IN(levels); CHECK_IN(Hact);
+    const int64_t B = levels.size(0), N = levels.size(1),
+                  L = levels.size(2), d = levels.size(3);
+    const int64_t G = dHpre.size(0), M = dHpre.size(1), m4 = dHpre.size(2);
+    auto opts = levels.options();
+    hipStream_t s = cur_stream();
+    torch::Tensor td_in;
+    if (mode == 1) {
+        auto pos = pos_opt.value();
+        CHECK_IN(pos);
+        td_in = torch::empty({B, N, G, d}, opts);
+        launch_add_pos(levels.data_ptr(), pos.data_ptr(), td_in.data_ptr(),
+                       td_in.numel(), (int)N, (int)L, (int)d, s);
+        check_launch();
+    }
+    auto dW1 = torch::empty({G * m4, d}, opts);
+    {
+        GemmParams p = base_params(m4, d, M, LAYOUT_TN, G, G, 1.0f);
+        p.A.base = dHpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
+        if (mode == 0) {
+            auto tokens = tokens_opt.value();
+            CHECK_IN(tokens);
+            p.B.flags = OP_TABLE;
+            p.Btab[0] = tokens.data_ptr();
+            p.Btabld[0] = d;
+            for (int64_t g = 1; g < G; g++) {
+                p.Btab[g] = (const char*)levels.data_ptr() + (g - 1) * d * 2;
+                p.Btabld[g] = L * d;
+            }
+        } else {
+            p.B.base = td_in.data_ptr();
+            p.B.sin = d; p.B.ld = G * d;
+        }
+        p.Cbase = dW1.data_ptr(); p.Csin = m4 * d; p.Cld = d;
+        run_gemm(p, s, opts, true);
+    }
+    auto dW2 = torch::empty({G * d, m4}, opts);
+    {
+        GemmParams p = base_params(d, m4, M, LAYOUT_TN, G, G, 1.0f);
+        p.A.base = (const char*)dY.data_ptr(); p.A.sin = d; p.A.ld = G * d;
+        p.B.base = Hact.data_ptr(); p.B.sin = M * m4; p.B.ld = m4;
+        p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
+        run_gemm(p, s, opts, true);
+    }
+    auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();
+    return {dW1, dW2, dB2};
+}
+
 std::string build_info() {
     return "glom_pytorch_amd HIP extension (gfx950, bf16 MFMA 16x16x32)";
 }
@@ -639,6 +759,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("glom_step_fwd", &glom_step_fwd, "full GLOM iteration forward");
     m.def("glom_step_bwd", &glom_step_bwd, "full GLOM iteration backward");
     m.def("add4_into", &add4_into, "fused 4-way elementwise sum");
+    m.def("ff_bwd_dh", &ff_bwd_dh);
+    m.def("ff_bwd_dx", &ff_bwd_dx);
+    m.def("ff_bwd_dw", &ff_bwd_dw);
     m.def("build_info", &build_info);
     m.def("bench_gemm", &bench_gemm, "raw GEMM microbench (tuning only)");
 }
