@@ -161,6 +161,35 @@ class GlomStepFn(torch.autograd.Function):
             return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
                     dtw1, dtb1, dtw2, dtb2, None, None, None, None, None,
                     None)
+        if os.environ.get("GLOM_BWD_FORK", "4") == "3":
+            cur = torch.cuda.current_stream()
+            s_td, s_at, _ = GlomStepFn._side_streams()
+            dmix, dtd = ext.level_mix_bwd(dnew.contiguous())
+            ev = torch.cuda.Event()
+            ev.record(cur)
+            bu = ext.grouped_ff_bwd(dmix, tokens, levels, None, bw1, bw2,
+                                    bhp, bha, 0, bw1t, bw2t)
+            with torch.cuda.stream(s_td):
+                s_td.wait_event(ev)
+                td = ext.grouped_ff_bwd(dtd, None, levels, pos, tw1, tw2,
+                                        thp, tha, 1, tw1t, tw2t)
+            with torch.cuda.stream(s_at):
+                s_at.wait_event(ev)
+                dAttn = ext.consensus_bwd(dmix, levels, probs, rnorm,
+                                          ctx.attend_self, mask)
+            dmix.record_stream(s_at)
+            dtd.record_stream(s_td)
+            cur.wait_stream(s_td)
+            cur.wait_stream(s_at)
+            for t in list(td) + [dAttn]:
+                if t is not None and t.numel():
+                    t.record_stream(cur)
+            dLevels = torch.empty_like(levels)
+            ext.add4_into(dmix, bu[1], td[1], dAttn, dLevels)
+            dPos = td[1][:, :, 1:, :].sum(dim=(0, 2))
+            return (bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
+                    td[2], td[3], td[4], td[5], None, None, None, None,
+                    None, None)
         # fork the independent backward work across four streams:
         #   cur : mix bwd -> bottom-up dH -> bottom-up dX
         #   s_td: top-down dH -> top-down dX
