@@ -161,7 +161,7 @@ class GlomStepFn(torch.autograd.Function):
             return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
                     dtw1, dtb1, dtw2, dtb2, None, None, None, None, None,
                     None)
-        if os.environ.get("GLOM_BWD_FORK", "4") == "3":
+        if os.environ.get("GLOM_BWD_FORK", "3") == "3":
             cur = torch.cuda.current_stream()
             s_td, s_at, _ = GlomStepFn._side_streams()
             dmix, dtd = ext.level_mix_bwd(dnew.contiguous())
@@ -190,13 +190,9 @@ class GlomStepFn(torch.autograd.Function):
             return (bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
                     td[2], td[3], td[4], td[5], None, None, None, None,
                     None, None)
-        # fork the independent backward work across four streams:
-        #   cur : mix bwd -> bottom-up dH -> bottom-up dX
-        #   s_td: top-down dH -> top-down dX
-        #   s_at: consensus attention backward
-        #   s_w : the four weight-grad GEMMs (depend only on the dH's)
-        # measured ~+7% (3-way) and more with the weight-grad offload;
-        # record_stream pins blocks that cross stream boundaries.
+        # 4-way fork variant (weight grads on a dedicated stream): measured
+        # slightly SLOWER than the 3-way fork above (sync overhead), kept
+        # behind GLOM_BWD_FORK=4 for future tuning.
         B, N, L = levels.size(0), levels.size(1), levels.size(2)
         cur = torch.cuda.current_stream()
         s_td, s_at, s_w = GlomStepFn._side_streams()
