@@ -339,3 +339,25 @@ def test_backward_singlestream_path_matches():
             assert ((a - b).norm() / b.norm().clamp_min(1e-8)) < 1e-2, n
         else:
             assert torch.equal(g_fork[n], g_single[n]), n
+
+
+def test_trainer_checkpoint_resume_gpu():
+    """Checkpoint/resume on GPU: identical weights and RNG-driven noise
+    give identical continuation losses."""
+    from glom_pytorch_amd.parallel.trainer import DenoisingTrainer
+    import tempfile, os
+    torch.manual_seed(0)
+    m = Glom(**CFG).to(DEV, torch.bfloat16)
+    tr = DenoisingTrainer(m, noise_std=0.5, decode_step=2)
+    img = torch.randn(2, 3, 32, 32, device=DEV, dtype=torch.bfloat16)
+    tr.step(img, iters=3)
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "ck.pt")
+        tr.save_checkpoint(path)
+        m2 = Glom(**CFG).to(DEV, torch.bfloat16)
+        tr2 = DenoisingTrainer(m2, noise_std=0.5, decode_step=2)
+        tr2.load_checkpoint(path)
+        la = tr.step(img, iters=3)
+        lb = tr2.step(img, iters=3)
+    assert tr2.step_idx == 2
+    assert abs(la - lb) < 1e-4, (la, lb)
