@@ -50,9 +50,17 @@ class DenoisingTrainer:
             self.ddp_dec = BucketedDDP(self.decoder, bucket_bytes)
             # rank-liveness probe: a hung peer becomes a clean abort
             self.heartbeat = Heartbeat(every_steps=50)
-        self.opt = torch.optim.AdamW(
-            list(model.parameters()) + list(self.decoder.parameters()),
-            lr=lr, foreach=True)
+        params = list(model.parameters()) + list(self.decoder.parameters())
+        # bf16 training keeps fp32 master weights: updates of size lr*grad
+        # would otherwise partially round away in bf16 parameter storage
+        self.master = None
+        if p.dtype == torch.bfloat16:
+            self.master = [q.detach().float().requires_grad_(False)
+                           for q in params]
+            self._params = params
+            self.opt = torch.optim.AdamW(self.master, lr=lr, foreach=True)
+        else:
+            self.opt = torch.optim.AdamW(params, lr=lr, foreach=True)
         self.step_idx = 0
         self.log_path = log_path
 
@@ -74,7 +82,16 @@ class DenoisingTrainer:
             self.ddp_model.finalize()
             self.ddp_dec.finalize()
             self.heartbeat.tick()
-        self.opt.step()
+        if self.master is not None:
+            with torch.no_grad():
+                for mw, q in zip(self.master, self._params):
+                    if q.grad is not None:
+                        mw.grad = q.grad.float()
+                self.opt.step()
+                for mw, q in zip(self.master, self._params):
+                    q.data.copy_(mw)
+        else:
+            self.opt.step()
         self.step_idx += 1
         return loss.item() if sync_loss else loss.detach()
 
@@ -94,6 +111,7 @@ class DenoisingTrainer:
             "model": self.model.state_dict(),
             "decoder": self.decoder.state_dict(),
             "optimizer": self.opt.state_dict(),
+            "master": self.master,
             "step": self.step_idx,
             "rng": torch.get_rng_state(),
             "cuda_rng": (torch.cuda.get_rng_state()
@@ -106,6 +124,9 @@ class DenoisingTrainer:
         self.model.load_state_dict(ckpt["model"])
         self.decoder.load_state_dict(ckpt["decoder"])
         self.opt.load_state_dict(ckpt["optimizer"])
+        if ckpt.get("master") is not None and self.master is not None:
+            for mw, saved in zip(self.master, ckpt["master"]):
+                mw.data.copy_(saved.to(mw.device))
         self.step_idx = ckpt["step"]
         torch.set_rng_state(ckpt["rng"])
         if ckpt.get("cuda_rng") is not None and torch.cuda.is_available():

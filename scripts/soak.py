@@ -1,19 +1,34 @@
-"""Training-stability soak: 200 denoising steps on the native engine; the
-loss must fall materially below its initial value and stay finite."""
-import sys, os, json, torch
+"""Training-stability soak: 300 denoising steps on structured synthetic
+images (low-frequency patterns are actually reconstructable from the top
+level); loss must fall materially and stay finite."""
+import sys, os, json, torch, math
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from glom_pytorch_amd import Glom
 from glom_pytorch_amd.parallel.trainer import DenoisingTrainer
 
 torch.manual_seed(0)
-m = Glom(dim=512, levels=6, image_size=224, patch_size=14).to("cuda", torch.bfloat16)
-tr = DenoisingTrainer(m, lr=1e-4, noise_std=0.3)
+dev = "cuda"
+
+def structured_batch(B, size=224):
+    # random low-frequency images: sum of a few 2-D sinusoids per channel
+    y, x = torch.meshgrid(torch.linspace(0, 1, size, device=dev),
+                          torch.linspace(0, 1, size, device=dev),
+                          indexing="ij")
+    img = torch.zeros(B, 3, size, size, device=dev)
+    for _ in range(4):
+        fx, fy = torch.randint(1, 5, (2,), device=dev)
+        ph = torch.rand(B, 3, 1, 1, device=dev) * 2 * math.pi
+        amp = torch.randn(B, 3, 1, 1, device=dev) * 0.5
+        img += amp * torch.sin(2 * math.pi * (fx * x + fy * y) + ph)
+    return img.to(torch.bfloat16)
+
+m = Glom(dim=512, levels=6, image_size=224, patch_size=14).to(dev, torch.bfloat16)
+tr = DenoisingTrainer(m, lr=1e-3, noise_std=0.3)
 losses = []
-for step in range(200):
-    img = torch.randn(32, 3, 224, 224, device="cuda", dtype=torch.bfloat16)
-    losses.append(tr.step(img, iters=12))
+for step in range(300):
+    losses.append(tr.step(structured_batch(32), iters=12))
 first, last = sum(losses[:10]) / 10, sum(losses[-10:]) / 10
-print(json.dumps({"first10": first, "last10": last,
-                  "min": min(losses), "finite": all(l == l for l in losses)}))
-assert last < 0.8 * first, (first, last)
+print(json.dumps({"first10": first, "last10": last, "min": min(losses),
+                  "finite": all(l == l for l in losses)}))
+assert last < 0.7 * first, (first, last)
 print("SOAK OK")
