@@ -34,7 +34,8 @@ class DenoisingDecoder(nn.Sequential):
 
 class DenoisingTrainer:
     def __init__(self, model, *, lr=3e-4, noise_std=1.0, decode_step=7,
-                 distributed=False, bucket_bytes=16 << 20, log_path=None):
+                 grad_clip=1.0, distributed=False, bucket_bytes=16 << 20,
+                 log_path=None):
         self.model = model
         self.dim = model.dim
         self.decode_step = decode_step
@@ -61,6 +62,7 @@ class DenoisingTrainer:
             self.opt = torch.optim.AdamW(self.master, lr=lr, foreach=True)
         else:
             self.opt = torch.optim.AdamW(params, lr=lr, foreach=True)
+        self.grad_clip = grad_clip
         self.step_idx = 0
         self.log_path = log_path
 
@@ -87,10 +89,17 @@ class DenoisingTrainer:
                 for mw, q in zip(self.master, self._params):
                     if q.grad is not None:
                         mw.grad = q.grad.float()
+                if self.grad_clip:
+                    torch.nn.utils.clip_grad_norm_(self.master,
+                                                   self.grad_clip)
                 self.opt.step()
                 for mw, q in zip(self.master, self._params):
                     q.data.copy_(mw)
         else:
+            if self.grad_clip:
+                torch.nn.utils.clip_grad_norm_(
+                    [q for g in self.opt.param_groups for q in g["params"]],
+                    self.grad_clip)
             self.opt.step()
         self.step_idx += 1
         return loss.item() if sync_loss else loss.detach()
