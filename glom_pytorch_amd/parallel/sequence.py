@@ -220,18 +220,38 @@ def sp_forward(model, img, iters: int | None = None,
         raise ValueError(f"levels has {levels.shape[1]} columns; expected "
                          f"{n} (full) or {n1 - n0} (local shard)")
 
+    # bf16-on-GPU ranks run their local column shard through the CDNA4
+    # grouped-GEMM / level-mix kernels (the FF work, ~3/4 of the FLOPs, is
+    # column-local); only the attention exchange stays in torch ops.
+    use_native = model._use_native(img)
+    if use_native:
+        from glom_pytorch_amd.ops.functional import GroupedFFFn, LevelMixFn
+        tokens_loc = tokens[:, n0:n1].contiguous()
+        pos_loc = model.pos_emb.weight[n0:n1].contiguous()
+        bw, tw = model.bottom_up.net, model.top_down.net
+        lv = lv.contiguous()
+
     contrib = model._contrib.to(lv.dtype)
     trajectory = [lv]
     for _ in range(iters):
-        bu_in = torch.cat((bottom, lv[..., :-1, :]), dim=-2)
-        bu = model.bottom_up(bu_in)
-        td = model.top_down(lv[..., 1:, :] + pos)
-        td = F.pad(td, (0, 0, 0, 1), value=0.0)
         if mode == "allgather":
             consensus = _attention_allgather(model, lv, n0, n1, pg)
         else:
             consensus = _attention_ring(model, lv, n0, n1, n, pg)
-        lv = (lv + bu + td + consensus) / contrib
+        if use_native:
+            bu = GroupedFFFn.apply(tokens_loc, lv, None,
+                                   bw[1].weight[..., 0], bw[1].bias,
+                                   bw[3].weight[..., 0], bw[3].bias, 0)
+            td = GroupedFFFn.apply(None, lv, pos_loc,
+                                   tw[1].weight[..., 0], tw[1].bias,
+                                   tw[3].weight[..., 0], tw[3].bias, 1)
+            lv = LevelMixFn.apply(lv, bu, td, consensus.contiguous())
+        else:
+            bu_in = torch.cat((bottom, lv[..., :-1, :]), dim=-2)
+            bu = model.bottom_up(bu_in)
+            td = model.top_down(lv[..., 1:, :] + pos)
+            td = F.pad(td, (0, 0, 0, 1), value=0.0)
+            lv = (lv + bu + td + consensus) / contrib
         trajectory.append(lv)
 
     out = torch.stack(trajectory) if return_all else lv
