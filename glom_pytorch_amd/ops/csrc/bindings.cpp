@@ -65,7 +65,31 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         }
     }
     const bool nt3 = nt_fast && p.N % 256 == 0 && p.N >= 1024;
-    if (nt3)
+    // persistent continuous-ring variant (default ON; opt out with
+    // GLOM_NT5P=0): one block sweeps 4 M-tiles of a column, register
+    // epilogue in the shadow of the next tile's MFMAs, fused colsum kept
+    static const bool nt5p_on = []() {
+        const char* e = getenv("GLOM_NT5P");
+        return !e || e[0] != '0';
+    }();
+    const bool nt5p = nt5p_on && nt3 && p.M % 512 == 0
+                      && !(p.Cflags & OP_TABLE)
+                      && (p.epilogue == EPI_NONE
+                          || p.epilogue == EPI_GELUGRAD);
+    static const bool disp_dbg = []() {
+        const char* e = getenv("GLOM_DISPATCH_DEBUG");
+        return e && e[0] == '1';
+    }();
+    if (disp_dbg)
+        fprintf(stderr,
+                "[dispatch] L%d M%ld N%ld K%ld epi%d nt3=%d nt5p=%d on=%d "
+                "ctabflag=%d m512=%d\n",
+                p.layout, (long)p.M, (long)p.N, (long)p.K, p.epilogue,
+                (int)nt3, (int)nt5p, (int)nt5p_on, (int)(p.Cflags & OP_TABLE),
+                (int)(p.M % 512 == 0));
+    if (nt5p)
+        launch_gemm_nt_fast5p(p, s);
+    else if (nt3)
         launch_gemm_nt_fast4(p, s);   // 3-ring counted-vmcnt variant
     else if (nt_fast)
         launch_gemm_nt_fast(p, s);

@@ -17,6 +17,8 @@
 // (TN: K%8==0 with row predicates), all row strides %8==0, no on-load
 // transforms. Everything else falls back to the generic gemm_kernel.
 
+#include <cstdlib>
+
 #include "common.h"
 #include "gemm.h"
 #include "gemm_device.h"
@@ -1172,6 +1174,242 @@ void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream) {
 }
 
 
+// ------- persistent continuous-ring NT variant (nt5p) -------
+// Each block sweeps PERSIST consecutive 128-row M-tiles of ONE 256-col
+// N-column. The 3-ring never drains at tile boundaries (the global step
+// stream g = tile*nk + kt staggers straight across tiles), so the per-tile
+// prologue cost — ~50% of a K=512 block in nt4 (profiles/README
+// K-amortization) — is paid once per PERSIST tiles. The epilogue runs from
+// registers (scalar stores; no LDS image — the ring owns the arena) in the
+// shadow of the next tile's MFMAs. Supports alpha/bias/colscale/GELUGRAD
+// and the fused f32 colsum (per-wave atomics, 2/column/tile); no
+// softmax/pair/table-C epilogues — the dispatcher routes those to nt4.
+template <int PERSIST>
+__global__ __launch_bounds__(NT3) void gemm_nt_fast5p_kernel(GemmParams p) {
+    __shared__ ushort_t smem[3 * (128 + 256) * FBK];   // 144 KiB, 3-ring
+    const int abuf = 128 * FBK, bbuf = 256 * FBK, stride = abuf + bbuf;
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN3;
+    const int m0 = (bid % gridDim.y) * (BM * PERSIST);
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    float* colp = nullptr;
+    if (p.colsum_out)
+        colp = p.colsum_out + (long)(pid % p.nInner) * p.colsum_sin;
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int wm = (wid >> 2) * 64;
+    const int wn = (wid & 3) * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    float csv[4] = {1.f, 1.f, 1.f, 1.f};
+    float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+    if (csp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+    }
+    if (biasp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+    }
+
+    f32x4 acc[4][4] = {};
+
+    const int nk = p.K / FBK;
+    const int TOT = PERSIST * nk;
+    auto stage_g = [&](int buf, int g) {
+        const int mt0 = m0 + (g / nk) * BM;
+        const int k0 = (g % nk) * FBK;
+        ushort_t* Al = smem + buf * stride;
+        ushort_t* Bl = Al + abuf;
+#pragma unroll
+        for (int c = 0; c < 2; c++) {
+            int chunk = wid * 2 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g_ = Ap + (long)(mt0 + row) * lda + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g_,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (Al + chunk * 512), 16, 0, 0);
+        }
+#pragma unroll
+        for (int c = 0; c < 4; c++) {
+            int chunk = wid * 4 + c;
+            int row = chunk * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g_ = Bp + (long)(n0 + row) * ldb + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g_,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (Bl + chunk * 512), 16, 0, 0);
+        }
+    };
+
+    stage_g(0, 0);
+    if (TOT > 1) stage_g(1, 1);
+    if (TOT > 1)
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (int g = 0; g < TOT; g++) {
+        const bool tile_end = (g % nk) == nk - 1;
+        // vmcnt retires IN ISSUE ORDER, so at tile-end steps the next
+        // stage's glds must be issued AFTER the epilogue stores: the
+        // bottom vmcnt(6) then covers [g+1 glds, stores] and leaves g+2's
+        // glds in flight. Stores retire once their data leaves the VGPRs
+        // (no memory round trip), so that wait stays cheap.
+        if (!tile_end && g + 2 < TOT) stage_g((g + 2) % 3, g + 2);
+        const ushort_t* Al = smem + (g % 3) * stride;
+        const ushort_t* Bl = Al + abuf;
+        short8 af[2][4], bfr[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++) {
+#pragma unroll
+            for (int i = 0; i < 4; i++) {
+                int row = wm + i * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                af[s][i] = *(const short8*)&Al[row * FBK + off];
+            }
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfr[s][j] = *(const short8*)&Bl[row * FBK + off];
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int i = 0; i < 4; i++)
+#pragma unroll
+                for (int j = 0; j < 4; j++)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[s][i], bfr[s][j], acc[i][j], 0, 0, 0);
+
+        if (tile_end) {
+            // tile done: drain it from registers while the ring streams the
+            // next tile
+            const long mt0 = m0 + (long)(g / nk) * BM;
+            float colacc[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int i16 = 0; i16 < 4; i16++) {
+#pragma unroll
+                for (int r = 0; r < 4; r++) {
+                    long gi = mt0 + wm + i16 * 16 + kq * 4 + r;
+                    float vv[4];
+#pragma unroll
+                    for (int j16 = 0; j16 < 4; j16++)
+                        vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+                    if (auxp) {
+                        const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                        ushort_t av[4];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            av[j16] = auxrow[n0 + wn + j16 * 16 + lrow];
+                        float gx[4], gy[4];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            gx[j16] = bf2f(av[j16]);
+                        gelu_grad_vec<4>(gx, gy);
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++) vv[j16] *= gy[j16];
+                    }
+                    ushort_t* crow = Cp + gi * ldc + n0 + wn + lrow;
+#pragma unroll
+                    for (int j16 = 0; j16 < 4; j16++) {
+                        float v = vv[j16] + bvv[j16];
+                        colacc[j16] += v;
+                        crow[j16 * 16] = f2bf(v);
+                    }
+                    acc[i16][0][r] = 0.f; acc[i16][1][r] = 0.f;
+                    acc[i16][2][r] = 0.f; acc[i16][3][r] = 0.f;
+                }
+            }
+            if (colp) {
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    atomicAdd(&colp[n0 + wn + j16 * 16 + lrow], colacc[j16]);
+            }
+            if (g + 2 < TOT) stage_g((g + 2) % 3, g + 2);
+        }
+
+        if (g + 1 < TOT) {
+            if (g + 2 < TOT)
+                asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+            else
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_s_barrier();
+        }
+    }
+}
+
+void launch_gemm_nt_fast5p(const GemmParams& p, hipStream_t stream) {
+    // sweep depth: 4 by default (dispatcher guarantees M % 512 == 0);
+    // GLOM_NT5P_P=8 selects the deeper sweep when M allows it
+    static const int pref = []() {
+        const char* e = getenv("GLOM_NT5P_P");
+        return e ? atoi(e) : 0;
+    }();
+    const int P = (pref == 8 && p.M % (BM * 8) == 0) ? 8
+                : (pref == 2) ? 2 : 4;
+    dim3 grid(p.N / BN3, p.M / (BM * P), p.nproblems);
+    if (P == 8)
+        hipLaunchKernelGGL(gemm_nt_fast5p_kernel<8>, grid, dim3(NT3), 0,
+                           stream, p);
+    else if (P == 2)
+        hipLaunchKernelGGL(gemm_nt_fast5p_kernel<2>, grid, dim3(NT3), 0,
+                           stream, p);
+    else
+        hipLaunchKernelGGL(gemm_nt_fast5p_kernel<4>, grid, dim3(NT3), 0,
+                           stream, p);
+}
+
+
 
 // ------- split-K-only TN variant (32 KiB arena, 4 blocks/CU) -------
 __global__ __launch_bounds__(NTHREADS) void gemm_tn_sk_kernel(GemmParams p) {
@@ -1284,4 +1522,6 @@ void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
 //  - 128x256 TN (68 KiB arena halves tn_fast's 4 blocks/CU: dW1 +37%)
 //  - 256^2 BK=32 3-ring (paired-row LDS; slower on K=512 and buggy)
 // The shipping set: gemm_nt_fast (128^2 glds 2-buf), gemm_nt_fast3/4
-// (128x256 8-wave, 2-buf / 3-ring), gemm_tn_fast(+_sk), gemm_nn_fast.
+// (128x256 8-wave, 2-buf / 3-ring), gemm_nt_fast5p (persistent
+// continuous-ring, default for M%512==0 plain/GELUGRAD epilogues),
+// gemm_tn_fast(+_sk), gemm_nn_fast.
