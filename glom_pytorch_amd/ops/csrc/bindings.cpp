@@ -72,7 +72,11 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         const char* e = getenv("GLOM_NT5P");
         return !e || e[0] != '0';
     }();
+    // N >= 1024 only: at N=512 (2 column panels) the persistent sweep
+    // measured SLOWER than the 128^2 kernel on the down-projection
+    // (637 -> 573 TF), so that shape keeps nt_fast
     const bool nt5p = nt5p_on && nt3 && p.M % 512 == 0
+                      && (long)(p.N / 256) * (p.M / 512) * p.nproblems >= 256
                       && !(p.Cflags & OP_TABLE)
                       && (p.epilogue == EPI_NONE
                           || p.epilogue == EPI_GELUGRAD);
