@@ -1514,6 +1514,7 @@ void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
     hipLaunchKernelGGL(gemm_tn_sk_kernel, grid, dim3(NTHREADS), 0, stream, p);
 }
 
+
 // ---------------------------------------------------------------- //
 // Measured-and-rejected variants (see git history + profiles/README.md):
 //  - 256^2 2-phase and phase-pipelined kernels (1 block/CU sync stalls
@@ -1521,6 +1522,11 @@ void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
 //  - 128^2 3-ring (96 KiB LDS halves occupancy: down 292->357us)
 //  - 128x256 TN (68 KiB arena halves tn_fast's 4 blocks/CU: dW1 +37%)
 //  - 256^2 BK=32 3-ring (paired-row LDS; slower on K=512 and buggy)
+//  - reg-pipelined TN 2-buf (issue-early/write-late, 64 KiB, 1 barrier
+//    per step): loses to tn_fast/tn_sk on every dW shape (486->477,
+//    560->542, 730->703, 775->756 TF; bench 962->943) — 4 blocks/CU
+//    cross-block overlap hides the repack latency better than in-kernel
+//    pipelining at 2 blocks/CU
 // The shipping set: gemm_nt_fast (128^2 glds 2-buf), gemm_nt_fast3/4
 // (128x256 8-wave, 2-buf / 3-ring), gemm_nt_fast5p (persistent
 // continuous-ring, default for M%512==0 plain/GELUGRAD epilogues),
