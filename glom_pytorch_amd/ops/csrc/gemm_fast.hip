@@ -1527,6 +1527,13 @@ void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream) {
 //    560->542, 730->703, 775->756 TF; bench 962->943) — 4 blocks/CU
 //    cross-block overlap hides the repack latency better than in-kernel
 //    pipelining at 2 blocks/CU
+//  - 128^2 persistent 3-ring (nt5p idea at BN=128, 96 KiB, 1 block/CU):
+//    4-wave/64x64 cut hit 260 VGPRs (1 wave/SIMD) and halved throughput;
+//    the 8-wave/64x32 cut (91 VGPRs) still lost on every target shape
+//    (down 622->464 TF, dP 457->407; bench 949->908) — at 128-wide
+//    tiles, 2 blocks/CU of the plain 2-buf kernel beats any 1-block/CU
+//    in-ring overlap. The wide-tile nt5p wins come from intensity AND
+//    persistence together, not persistence alone
 // The shipping set: gemm_nt_fast (128^2 glds 2-buf), gemm_nt_fast3/4
 // (128x256 8-wave, 2-buf / 3-ring), gemm_nt_fast5p (persistent
 // continuous-ring, default for M%512==0 plain/GELUGRAD epilogues),
