@@ -91,3 +91,20 @@ def test_consensus_fwd_bwd_shapes(B, N, L, d):
     (dref,) = torch.autograd.grad(ref, x, gout.float())
     assert _rel(out, ref) < 1.5e-2
     assert _rel(dlev, dref) < 5e-2
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_nt5p_dispatches_on_headline_shapes():
+    """The persistent-ring kernel must actually own the up/dH shapes —
+    guards against a dispatcher regression silently rerouting to nt4."""
+    import os
+    import subprocess
+    import sys
+    env = dict(os.environ, GLOM_DISPATCH_DEBUG="1")
+    code = ("import torch; from glom_pytorch_amd.ops import _load_extension;"
+            " ext=_load_extension(); ext.bench_gemm(16384,2048,512,0,6,0,1)")
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "nt5p=1" in r.stderr, r.stderr[-800:]
