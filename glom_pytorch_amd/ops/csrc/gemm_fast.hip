@@ -1362,9 +1362,13 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast5p_kernel(GemmParams p) {
                     ushort_t* crow = Cp + gi * ldc + n0 + wn + lrow;
 #pragma unroll
                     for (int j16 = 0; j16 < 4; j16++) {
-                        float v = vv[j16] + bvv[j16];
-                        colacc[j16] += v;
-                        crow[j16 * 16] = f2bf(v);
+                        ushort_t cb = f2bf(vv[j16] + bvv[j16]);
+                        // colsum over the bf16-ROUNDED stored values,
+                        // matching the LDS-epilogue colsum bit for bit
+                        // per summand (only the add order differs, and
+                        // that was already atomic-nondeterministic)
+                        colacc[j16] += bf2f(cb);
+                        crow[j16 * 16] = cb;
                     }
                     acc[i16][0][r] = 0.f; acc[i16][1][r] = 0.f;
                     acc[i16][2][r] = 0.f; acc[i16][3][r] = 0.f;
