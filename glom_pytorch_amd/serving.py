@@ -108,6 +108,12 @@ def create_app(model, iters: int = 12, max_batch: int = 16,
     @app.post("/embed")
     async def embed(request: Request):
         body = await request.body()
+        expected = 3 * size * size * 4
+        if len(body) != expected:
+            return Response(
+                status_code=400,
+                content=(f"expected {expected} bytes "
+                         f"(f32 3x{size}x{size}), got {len(body)}"))
         img = np.frombuffer(body, dtype=np.float32).reshape(3, size, size)
         out = await batcher.submit(torch.from_numpy(img.copy()))
         return Response(content=out.numpy().tobytes(),

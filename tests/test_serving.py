@@ -70,3 +70,12 @@ def test_embed_endpoint_gpu_graphs():
             ref = model(torch.from_numpy(img)[None].to("cuda",
                         torch.bfloat16), iters=3)[0, :, -1]
         assert np.allclose(out, ref.float().cpu().numpy(), atol=1e-2)
+
+
+def test_embed_rejects_malformed_body():
+    model = Glom(**SMALL)
+    app = create_app(model, iters=2)
+    with TestClient(app) as client:
+        r = client.post("/embed", content=b"\x00" * 100)
+        assert r.status_code == 400
+        assert b"expected" in r.content
