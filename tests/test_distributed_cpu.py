@@ -96,3 +96,16 @@ def test_trainer_checkpoint_roundtrip(tmp_path):
     la = tr.step(img, iters=3)
     lb = tr2.step(img, iters=3)
     assert abs(la - lb) < 1e-6
+
+
+def test_trainer_deferred_loss_sync():
+    """sync_loss=False returns the loss as a detached tensor (no host
+    sync) — the bench path; values must agree with the synced path."""
+    from glom_pytorch_amd.parallel.trainer import DenoisingTrainer
+    torch.manual_seed(0)
+    model = Glom(**SMALL)
+    tr = DenoisingTrainer(model, noise_std=0.5, decode_step=2)
+    img = torch.randn(2, 3, 32, 32)
+    t = tr.step(img, iters=2, sync_loss=False)
+    assert torch.is_tensor(t) and not t.requires_grad
+    assert t.item() > 0
