@@ -164,6 +164,13 @@ class Glom(nn.Module):
 
     # ------------------------------------------------------------------ #
 
+    def _native_capable(self) -> bool:
+        """Configs the HIP engine covers: dim a multiple of 8 (16B bf16
+        vector loads) and <= 16 levels (the grouped-GEMM pointer-table
+        width). Everything the reference README exercises qualifies;
+        exotic configs run the eager path with a one-time warning."""
+        return self.dim % 8 == 0 and self.levels <= 16
+
     def _use_native(self, img: torch.Tensor) -> bool:
         if not img.is_cuda:
             return False
@@ -171,7 +178,18 @@ class Glom(nn.Module):
             return False
         if getattr(self, "force_eager", False):
             return False
-        return img.dtype == torch.bfloat16
+        if img.dtype != torch.bfloat16:
+            return False
+        if not self._native_capable():
+            if not getattr(self, "_warned_capability", False):
+                import warnings
+                warnings.warn(
+                    f"Glom(dim={self.dim}, levels={self.levels}) is outside "
+                    "the HIP engine's envelope (dim%8==0, levels<=16); "
+                    "running the eager path on GPU.")
+                self._warned_capability = True
+            return False
+        return True
 
     def forward(self, img: torch.Tensor, iters: int | None = None,
                 levels: torch.Tensor | None = None,

@@ -132,3 +132,19 @@ def test_fuzz_configs_vs_oracle():
         out = m(img, iters=it)
         ref = oracle_forward(m, img, iters=it)
         assert torch.allclose(out, ref, rtol=5e-5, atol=5e-5), (kw, it)
+
+
+def test_native_capability_envelope():
+    """Exotic configs (dim % 8 != 0, levels > 16) are outside the HIP
+    engine's envelope and must route to eager instead of crashing."""
+    m = Glom(dim=100, levels=3, image_size=32, patch_size=8)
+    assert not m._native_capable()
+    m2 = Glom(dim=64, levels=17, image_size=32, patch_size=8)
+    assert not m2._native_capable()
+    m3 = Glom(dim=64, levels=3, image_size=32, patch_size=8)
+    assert m3._native_capable()
+    # CPU forward works for all of them (eager path)
+    img = torch.randn(1, 3, 32, 32)
+    for mod in (m, m2, m3):
+        out = mod(img, iters=1)
+        assert out.shape == (1, 16, mod.levels, mod.dim)
