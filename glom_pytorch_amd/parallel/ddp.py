@@ -77,6 +77,19 @@ class BucketedDDP:
 
         Call between loss.backward() and optimizer.step().
         """
+        # Flush partially-filled buckets: a bucket never fires its hook
+        # count if one of its params got no gradient this step (e.g.
+        # init_levels when training the stateful path with `levels=`
+        # provided). Grad presence is structural — identical across DP
+        # ranks running the same step — so every rank flushes the same
+        # buckets in the same order and the collective order stays
+        # consistent.
+        for bi, b in enumerate(self.buckets):
+            if 0 < self._pending[bi] < len(b):
+                for p in b:
+                    if p.grad is None:
+                        p.grad = torch.zeros_like(p)
+                self._launch(bi)
         for w in self._works:
             w.wait()
         inv = 1.0 / self.world

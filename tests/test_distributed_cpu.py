@@ -109,3 +109,59 @@ def test_trainer_deferred_loss_sync():
     t = tr.step(img, iters=2, sync_loss=False)
     assert torch.is_tensor(t) and not t.requires_grad
     assert t.item() > 0
+
+
+def _worker_partial(rank, world, port, img_all, out_path):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)
+    from glom_pytorch_amd.parallel.ddp import BucketedDDP
+    model = Glom(**SMALL)
+    ddp = BucketedDDP(model, bucket_bytes=1 << 30)   # ONE bucket
+    shard = img_all.chunk(world)[rank]
+    # stateful path: init_levels gets NO grad, so the single bucket is
+    # partially filled and must be flushed by finalize()
+    with torch.no_grad():
+        warm = model(shard, iters=1)
+    out = model(shard, iters=2, levels=warm)
+    out[:, :, -1].pow(2).mean().backward()
+    assert model.init_levels.grad is None
+    ddp.finalize()
+    if rank == 0:
+        torch.save({n: (p.grad.clone() if p.grad is not None else None)
+                    for n, p in model.named_parameters()}, out_path)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_partial_bucket_stateful_path(tmp_path):
+    """A bucket with a no-grad param (init_levels under levels=) must
+    still all-reduce the params that DO have grads."""
+    torch.manual_seed(42)
+    img_all = torch.randn(4, 3, 32, 32)
+    out_path = str(tmp_path / "pg.pt")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_partial,
+                         args=(r, 2, 29572, img_all, out_path))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    got = torch.load(out_path, weights_only=False)
+
+    # single-process reference over the full batch
+    torch.manual_seed(0)
+    model = Glom(**SMALL)
+    with torch.no_grad():
+        warm = model(img_all, iters=1)
+    out = model(img_all, iters=2, levels=warm)
+    out[:, :, -1].pow(2).mean().backward()
+    for n, p in model.named_parameters():
+        if n == "init_levels":
+            # flushed as zeros on ranks; single-process has None
+            assert got[n] is None or torch.count_nonzero(got[n]) == 0
+            continue
+        assert torch.allclose(got[n], p.grad, rtol=1e-4, atol=1e-6), n
