@@ -87,8 +87,9 @@ class DenoisingTrainer:
         if self.master is not None:
             with torch.no_grad():
                 for mw, q in zip(self.master, self._params):
-                    if q.grad is not None:
-                        mw.grad = q.grad.float()
+                    # None (not stale) when the param got no grad this
+                    # step, e.g. init_levels on the stateful path
+                    mw.grad = q.grad.float() if q.grad is not None else None
                 if self.grad_clip:
                     torch.nn.utils.clip_grad_norm_(self.master,
                                                    self.grad_clip)
