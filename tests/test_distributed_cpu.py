@@ -11,6 +11,7 @@ import torch.multiprocessing as mp
 
 from glom_pytorch_amd import Glom
 from conftest import SMALL
+from _netutil import free_port
 
 
 def _grads_single(img_all):
@@ -49,7 +50,7 @@ def test_ddp_grad_parity_vs_single_process(tmp_path):
 
     out_path = str(tmp_path / "grads.pt")
     ctx = mp.get_context("spawn")
-    port = 29537
+    port = free_port()
     procs = [ctx.Process(target=_worker, args=(r, 2, port, img_all, out_path))
              for r in range(2)]
     for p in procs:
@@ -141,9 +142,10 @@ def test_ddp_partial_bucket_stateful_path(tmp_path):
     torch.manual_seed(42)
     img_all = torch.randn(4, 3, 32, 32)
     out_path = str(tmp_path / "pg.pt")
+    _pp = free_port()
     ctx = mp.get_context("spawn")
     procs = [ctx.Process(target=_worker_partial,
-                         args=(r, 2, 29572, img_all, out_path))
+                         args=(r, 2, _pp, img_all, out_path))
              for r in range(2)]
     for p in procs:
         p.start()

@@ -13,6 +13,7 @@ import torch.multiprocessing as mp
 
 from glom_pytorch_amd import Glom
 from conftest import SMALL
+from _netutil import free_port
 
 
 def _worker(rank, world, port, cfg, img, out_path):
@@ -76,7 +77,7 @@ def test_sp_forward_parity(tmp_path, radius):
         cfg["local_consensus_radius"] = radius
     torch.manual_seed(42)
     img = torch.randn(2, 3, 32, 32)
-    got = _run_world(cfg, img, tmp_path, 29561 + radius)
+    got = _run_world(cfg, img, tmp_path, free_port())
 
     torch.manual_seed(0)
     model = Glom(**cfg)
@@ -114,7 +115,7 @@ def test_sp_native_gpu_world1():
     kernels (GroupedFFFn/LevelMixFn); world-size-1 RCCL makes the gather
     trivial so the result must match the fused native forward."""
     os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = "29591"
+    os.environ["MASTER_PORT"] = str(free_port())
     dist.init_process_group("nccl", rank=0, world_size=1)
     try:
         from glom_pytorch_amd.parallel.sequence import sp_forward
