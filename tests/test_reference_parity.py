@@ -58,3 +58,26 @@ def test_stateful_matches_upstream():
     la = ours(img2, iters=2, levels=ours(img1, iters=3))
     lb = ref(img2, iters=2, levels=ref(img1, iters=3))
     assert torch.allclose(la, lb, rtol=1e-5, atol=1e-6)
+
+
+def test_gradients_match_upstream():
+    kw = dict(dim=64, levels=3, image_size=32, patch_size=8)
+    torch.manual_seed(3)
+    ours = Glom(**kw)
+    ref = _ref_glom(**kw)
+    ref.load_state_dict(ours.state_dict(), strict=True)
+    img = torch.randn(2, 3, 32, 32)
+
+    out_a = ours(img, iters=3, return_all=True)
+    out_b = ref(img, iters=3, return_all=True)
+    # loss over the full trajectory reaches every parameter
+    loss_a = out_a.pow(2).mean()
+    loss_b = out_b.pow(2).mean()
+    loss_a.backward()
+    loss_b.backward()
+    ref_named = dict(ref.named_parameters())
+    for n, p in ours.named_parameters():
+        q = ref_named[n]
+        assert p.grad is not None and q.grad is not None, n
+        assert torch.allclose(p.grad, q.grad, rtol=1e-4, atol=1e-7), \
+            (n, (p.grad - q.grad).abs().max().item())
