@@ -86,19 +86,20 @@ class MicroBatcher:
 
 def create_app(model, iters: int = 12, max_batch: int = 16,
                max_wait_ms: float = 2.0):
+    from contextlib import asynccontextmanager
+
     from fastapi import FastAPI, Request, Response
 
-    app = FastAPI(title="glom_pytorch_amd")
     batcher = MicroBatcher(model, iters, max_batch, max_wait_ms)
     size = model.image_size
 
-    @app.on_event("startup")
-    async def _start():
+    @asynccontextmanager
+    async def _lifespan(app):
         batcher.start()
-
-    @app.on_event("shutdown")
-    async def _stop():
+        yield
         await batcher.stop()
+
+    app = FastAPI(title="glom_pytorch_amd", lifespan=_lifespan)
 
     @app.get("/healthz")
     async def healthz():
