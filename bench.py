@@ -35,6 +35,9 @@ def parse_args():
                    default="native",
                    help="native = CDNA4 HIP engine; eager/compile = stock "
                         "PyTorch-ROCm running the same math (comparison)")
+    p.add_argument("--bucket-mb", type=int, default=16,
+                   help="DP gradient all-reduce bucket size (MB); tune "
+                        "against the 7-link xGMI ring bandwidth")
     p.add_argument("--mode", choices=["train", "infer"], default="train",
                    help="train = denoising fwd+bwd+AdamW (headline); "
                         "infer = no-grad forward under hipGraph replay")
@@ -86,7 +89,8 @@ def main():
                 model(x, iters=iters)
             return 0.0
     else:
-        trainer = DenoisingTrainer(model, distributed=distributed)
+        trainer = DenoisingTrainer(model, distributed=distributed,
+                                   bucket_bytes=args.bucket_mb << 20)
 
         def step(x, iters):
             # loss stays on-device inside the timed loop (no host sync);

@@ -83,9 +83,10 @@ class LevelMixFn(torch.autograd.Function):
     def backward(ctx, dout):
         ext = _load_extension()
         dmix, dtd = ext.level_mix_bwd(dout.contiguous())
-        # prev/bu/cons share the same scaled gradient; consumers only read
-        # their incoming grads, so aliasing one tensor three times is safe.
-        return dmix, dmix, dtd, dmix
+        # prev/bu/cons share the same scaled gradient; hand the engine
+        # distinct view objects so it can never steal/accumulate into one
+        # buffer in place on behalf of another input.
+        return dmix, dmix.view_as(dmix), dtd, dmix.view_as(dmix)
 
 
 class GlomStepFn(torch.autograd.Function):

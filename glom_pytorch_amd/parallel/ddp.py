@@ -9,6 +9,16 @@ gradients with the backward of earlier iterations.
 
 GLOM's 23.5 M params (47 MB bf16) fit in ~3 buckets; reverse registration
 order approximates backward completion order.
+
+Comm-stream ordering: torch's ProcessGroupNCCL (RCCL on ROCm) launches
+every collective on its own dedicated per-device internal stream, after
+inserting an event-wait on the producer (current) stream; `async_op=True`
+returns immediately and `work.wait()` only enqueues an event-wait on the
+caller's stream. The bucket all-reduces launched from the backward hooks
+therefore already run concurrently with the remaining backward kernels on
+the compute stream — no extra comm stream is needed at this layer. RCCL
+channel/ring tuning for the 7-link xGMI topology lives in
+`rccl_env.apply_rccl_env_defaults()` (applied by `init_distributed`).
 """
 
 from __future__ import annotations
@@ -60,6 +70,18 @@ class BucketedDDP:
 
     def _on_grad(self, p):
         bi = self._bucket_of[p]
+        # Exactly one backward per finalize() is supported: a second
+        # backward before finalize() would re-reduce stale flats and
+        # silently drop the extra microbatch's gradients. Fail loudly
+        # instead (gradient accumulation callers: accumulate locally and
+        # call finalize() once, or wrap microbatches in no_sync()-style
+        # logic of their own).
+        if self._pending[bi] <= 0:
+            raise RuntimeError(
+                "BucketedDDP saw a gradient hook fire after its bucket was "
+                "already launched: more than one backward() ran before "
+                "finalize(). BucketedDDP supports exactly one backward per "
+                "finalize().")
         self._pending[bi] -= 1
         if self._pending[bi] == 0:
             self._launch(bi)

@@ -17,6 +17,9 @@ def init_distributed(timeout_s: int = 120) -> tuple[int, int]:
     into a clean abort-and-relaunch instead of a wedged job. Returns
     (rank, world_size)."""
     if not dist.is_initialized():
+        from glom_pytorch_amd.parallel.rccl_env import (
+            apply_rccl_env_defaults)
+        apply_rccl_env_defaults()
         dist.init_process_group(
             "nccl", timeout=datetime.timedelta(seconds=timeout_s))
     return dist.get_rank(), dist.get_world_size()

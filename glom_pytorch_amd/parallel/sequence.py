@@ -12,7 +12,11 @@ Two attention exchange modes:
 - ``mode="allgather"`` (default, differentiable): each rank all-gathers the
   full ``levels`` tensor once per iteration (via
   ``torch.distributed.nn.functional.all_gather``, so gradients flow back and
-  SP *training* composes with the DP trainer), then computes attention for
+  SP *training* works; composing with the DP trainer for a 2-D DP x SP mesh
+  additionally needs the DP subgroup passed as ``DenoisingTrainer``'s
+  ``process_group`` — BucketedDDP averages over its group, while SP shards
+  of one replica need their weight grads SUM-reduced over the SP subgroup,
+  as tests/test_sequence_parallel_cpu.py does), then computes attention for
   its local query rows only. One bucketed collective per iteration —
   the RCCL/xGMI-friendly shape: ring all-gather over 7 point-to-point links,
   each rank receives (W-1)/W of B·N·L·d once.

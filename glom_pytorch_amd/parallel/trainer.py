@@ -35,7 +35,12 @@ class DenoisingDecoder(nn.Sequential):
 class DenoisingTrainer:
     def __init__(self, model, *, lr=3e-4, noise_std=1.0, decode_step=7,
                  grad_clip=1.0, distributed=False, bucket_bytes=16 << 20,
-                 log_path=None):
+                 process_group=None, log_path=None):
+        """``process_group``: the group whose ranks are data-parallel
+        replicas (default: the global group). For 2-D DP x SP meshes pass
+        the DP subgroup here — BucketedDDP AVERAGES over its group, which
+        is correct for DP replicas; SP shards need a separate SUM
+        all-reduce over the SP subgroup (see parallel/sequence.py)."""
         self.model = model
         self.dim = model.dim
         self.decode_step = decode_step
@@ -47,8 +52,10 @@ class DenoisingTrainer:
         self.distributed = distributed and dist.is_initialized()
         self.ddp_model = self.ddp_dec = self.heartbeat = None
         if self.distributed:
-            self.ddp_model = BucketedDDP(model, bucket_bytes)
-            self.ddp_dec = BucketedDDP(self.decoder, bucket_bytes)
+            self.ddp_model = BucketedDDP(model, bucket_bytes,
+                                         process_group=process_group)
+            self.ddp_dec = BucketedDDP(self.decoder, bucket_bytes,
+                                       process_group=process_group)
             # rank-liveness probe: a hung peer becomes a clean abort
             self.heartbeat = Heartbeat(every_steps=50)
         params = list(model.parameters()) + list(self.decoder.parameters())

@@ -14,6 +14,7 @@ liveness and queue depth.
 """
 
 import asyncio
+import concurrent.futures
 
 import numpy as np
 import torch
@@ -30,8 +31,15 @@ class MicroBatcher:
         self.max_wait = max_wait_ms / 1e3
         self.queue: asyncio.Queue = asyncio.Queue()
         self._task = None
+        # single worker thread: keeps the event loop responsive during the
+        # GPU forward + device sync while preserving batch ordering
+        self._pool = concurrent.futures.ThreadPoolExecutor(max_workers=1)
         p = next(model.parameters())
         self.device, self.dtype = p.device, p.dtype
+        # fixed padded batch shapes -> every batch replays a captured graph
+        if (p.is_cuda and hasattr(model, "enable_graphs")
+                and getattr(model, "_graph_cache", None) is None):
+            model.enable_graphs()
 
     def start(self):
         self._task = asyncio.get_event_loop().create_task(self._loop())
@@ -39,6 +47,7 @@ class MicroBatcher:
     async def stop(self):
         if self._task:
             self._task.cancel()
+        self._pool.shutdown(wait=False)
 
     async def submit(self, img: torch.Tensor) -> torch.Tensor:
         fut = asyncio.get_event_loop().create_future()
@@ -60,7 +69,8 @@ class MicroBatcher:
                 except asyncio.TimeoutError:
                     break
             try:
-                outs = self._run([b[0] for b in batch])
+                outs = await asyncio.get_event_loop().run_in_executor(
+                    self._pool, self._run, [b[0] for b in batch])
                 for (_, f), out in zip(batch, outs):
                     if not f.done():
                         f.set_result(out)
