@@ -14,8 +14,24 @@
 
 #include "gemm.h"
 #include "aux_kernels.h"
+#include "native_ops.h"
 
 namespace {
+
+// Untracked alias of `base` (shares storage, keeps it alive, but has its
+// own TensorImpl/version counter). Used for trajectory-slab slices: a
+// custom Function may return such an alias even though later steps write
+// other slices of the same slab — a torch view would trip the
+// view+inplace guard.
+torch::Tensor alias_slab_slice(const torch::Tensor& slab, int64_t idx,
+                               c10::IntArrayRef sizes) {
+    int64_t n = 1;
+    for (auto s : sizes) n *= s;
+    void* ptr = (char*)slab.data_ptr() + idx * n * slab.element_size();
+    auto keep = slab;   // captured by the deleter -> storage stays alive
+    return torch::from_blob(
+        ptr, sizes, [keep](void*) mutable {}, slab.options());
+}
 
 #define CHECK_IN(x)                                                        \
     TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                        \
@@ -220,7 +236,13 @@ std::vector<torch::Tensor> grouped_ff_bwd(
     hipStream_t s = cur_stream();
 
     auto dHpre = torch::empty({G, M, m4}, opts);
-    auto dLevels = torch::zeros({B, N, L, d}, opts);
+    // the scattered dX GEMM writes every level slice except one (mode 0:
+    // the top slice, which only the top-down path feeds; mode 1: slice 0);
+    // zero just that slice instead of a full-tensor fill
+    auto dLevels = torch::empty({B, N, L, d}, opts);
+    launch_zero_slice(dLevels.data_ptr(), B * N, (int)L, (int)d,
+                      mode == 0 ? (int)L - 1 : 0, s);
+    check_launch();
     torch::Tensor dTokens;
 
     // per-group weight transposes turn the NN data grads into NT GEMMs
@@ -313,9 +335,27 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
         run_gemm(p, s, opts, true);
     }
-    auto dB1 = dh_nt3 ? db1f.flatten().to(at::kBFloat16)
-                      : dHpre.sum(1).flatten();              // (G*m4)
-    auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();       // (G*d)
+    torch::Tensor dB1;
+    if (dh_nt3) {
+        dB1 = db1f.flatten().to(at::kBFloat16);
+    } else {
+        // native deterministic column sum (replaces ATen .sum(1))
+        dB1 = torch::empty({G * m4}, opts);
+        auto ws = torch::empty({(long)G * colsum_rb(M) * m4},
+                               opts.dtype(at::kFloat));
+        launch_colsum(dHpre.data_ptr(), ws.data_ptr<float>(),
+                      dB1.data_ptr(), (int)G, M, m4, s);
+        check_launch();
+    }
+    // dB2 = colsum over the M token rows of dY viewed as (M, G*d)
+    auto dB2 = torch::empty({G * d}, opts);
+    {
+        auto ws = torch::empty({(long)colsum_rb(M) * G * d},
+                               opts.dtype(at::kFloat));
+        launch_colsum(dY.data_ptr(), ws.data_ptr<float>(), dB2.data_ptr(),
+                      1, M, G * d, s);
+        check_launch();
+    }
     if (mode != 0) dTokens = torch::empty({0}, opts);
     return {dTokens, dLevels, dW1, dB1, dW2, dB2};
 }
@@ -508,12 +548,27 @@ torch::Tensor consensus_bwd(torch::Tensor dOut, torch::Tensor levels,
 // ------------------------------------------------------------------ //
 // level mixing (reference glom_pytorch.py:128-144)
 
+// When (slab, slab_idx) is given, the output is written straight into
+// slab[slab_idx] and returned as an untracked alias — the return_all
+// trajectory is then materialized in place, with no torch.stack copy
+// (SURVEY.md §2.3 note on the trajectory slab).
 torch::Tensor level_mix_fwd(torch::Tensor levels, torch::Tensor bu,
-                            torch::Tensor td, torch::Tensor cons) {
+                            torch::Tensor td, torch::Tensor cons,
+                            c10::optional<torch::Tensor> slab = c10::nullopt,
+                            int64_t slab_idx = 0) {
     CHECK_IN(levels); CHECK_IN(bu); CHECK_IN(td); CHECK_IN(cons);
     const int64_t L = levels.size(2), d = levels.size(3);
     TORCH_CHECK(d % 8 == 0, "dim must be a multiple of 8");
-    auto out = torch::empty_like(levels);
+    torch::Tensor out;
+    if (slab.has_value()) {
+        TORCH_CHECK(slab->is_contiguous()
+                    && slab->scalar_type() == at::kBFloat16
+                    && slab->numel() >= (slab_idx + 1) * levels.numel(),
+                    "bad trajectory slab");
+        out = alias_slab_slice(slab.value(), slab_idx, levels.sizes());
+    } else {
+        out = torch::empty_like(levels);
+    }
     launch_mix_fwd(levels.data_ptr(), bu.data_ptr(), td.data_ptr(),
                    cons.data_ptr(), out.data_ptr(), levels.numel(), (int)L,
                    (int)d, cur_stream());
@@ -547,13 +602,15 @@ std::vector<torch::Tensor> glom_step_fwd(
         torch::Tensor bw1, torch::Tensor bb1, torch::Tensor bw2,
         torch::Tensor bb2, torch::Tensor tw1, torch::Tensor tb1,
         torch::Tensor tw2, torch::Tensor tb2, bool attend_self,
-        c10::optional<torch::Tensor> mask) {
+        c10::optional<torch::Tensor> mask,
+        c10::optional<torch::Tensor> slab = c10::nullopt,
+        int64_t slab_idx = 0) {
     auto bu = grouped_ff_fwd(tokens, levels, c10::nullopt, bw1, bb1, bw2,
                              bb2, 0);
     auto td = grouped_ff_fwd(c10::nullopt, levels, pos, tw1, tb1, tw2,
                              tb2, 1);
     auto at = consensus_fwd(levels, attend_self, mask);
-    auto out = level_mix_fwd(levels, bu[0], td[0], at[0]);
+    auto out = level_mix_fwd(levels, bu[0], td[0], at[0], slab, slab_idx);
     return {out, bu[1], bu[2], td[1], td[2], at[1], at[2]};
 }
 
@@ -580,8 +637,14 @@ std::vector<torch::Tensor> glom_step_bwd(
                 dAttn.data_ptr(), dLevels.data_ptr(), levels.numel(),
                 cur_stream());
     check_launch();
-    auto dPos = td[1].narrow(2, 1, L - 1)
-                    .sum(std::vector<int64_t>{0, 2});   // (N, d)
+    // dPos[n,:] = sum over batch and levels 1..L-1 of the top-down input
+    // grads (native deterministic reduction, reference glom_pytorch.py:136)
+    const int64_t B = levels.size(0), N = levels.size(1),
+                  d = levels.size(3);
+    auto dPos = torch::empty({N, d}, levels.options());
+    launch_dpos(td[1].data_ptr(), dPos.data_ptr(), (int)B, (int)N, (int)L,
+                (int)d, cur_stream());
+    check_launch();
     return {bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
             td[2], td[3], td[4], td[5]};
 }
@@ -674,8 +737,17 @@ std::vector<torch::Tensor> ff_bwd_dh(torch::Tensor dY, torch::Tensor w2t,
         p.colsum_sin = m4;
     }
     run_gemm(p, s, opts, true);
-    auto dB1 = fused ? db1f.flatten().to(at::kBFloat16)
-                     : dHpre.sum(1).flatten();
+    torch::Tensor dB1;
+    if (fused) {
+        dB1 = db1f.flatten().to(at::kBFloat16);
+    } else {
+        dB1 = torch::empty({G * m4}, opts);
+        auto cws = torch::empty({(long)G * colsum_rb(M) * m4},
+                                opts.dtype(at::kFloat));
+        launch_colsum(dHpre.data_ptr(), cws.data_ptr<float>(),
+                      dB1.data_ptr(), (int)G, M, m4, s);
+        check_launch();
+    }
     return {dHpre, dB1};
 }
 
@@ -689,7 +761,10 @@ std::vector<torch::Tensor> ff_bwd_dx(torch::Tensor dHpre, torch::Tensor w1t,
     auto opts = dHpre.options();
     hipStream_t s = cur_stream();
     torch::Tensor dTokens;
-    auto dLevels = torch::zeros({B, N, L, d}, opts);
+    auto dLevels = torch::empty({B, N, L, d}, opts);
+    launch_zero_slice(dLevels.data_ptr(), B * N, (int)L, (int)d,
+                      mode == 0 ? (int)L - 1 : 0, s);
+    check_launch();
     GemmParams p = base_params(M, d, m4, LAYOUT_NT, G, G, 1.0f);
     p.A.base = dHpre.data_ptr(); p.A.sin = M * m4; p.A.ld = m4;
     p.B.base = w1t.data_ptr(); p.B.sin = d * m4; p.B.ld = m4;
@@ -762,8 +837,174 @@ std::vector<torch::Tensor> ff_bwd_dw(
         p.Cbase = dW2.data_ptr(); p.Csin = d * m4; p.Cld = m4;
         run_gemm(p, s, opts, true);
     }
-    auto dB2 = dY.reshape({M, G, d}).sum(0).flatten();
+    auto dB2 = torch::empty({G * d}, opts);
+    {
+        auto cws = torch::empty({(long)colsum_rb(M) * G * d},
+                                opts.dtype(at::kFloat));
+        launch_colsum(dY.data_ptr(), cws.data_ptr<float>(), dB2.data_ptr(),
+                      1, M, G * d, s);
+        check_launch();
+    }
     return {dW1, dW2, dB2};
+}
+
+// ------------------------------------------------------------------ //
+// patch embedding (reference glom_pytorch.py:94-97,114): patchify
+// rearrange + Linear(p^2*3 -> dim), K-padded to a /64 boundary so the
+// tuned NT kernel serves the GEMM (zero columns are exact no-ops).
+
+static int64_t pad_k(int64_t k) { return (k + 63) / 64 * 64; }
+
+std::vector<torch::Tensor> patch_embed_fwd(torch::Tensor img,
+                                           torch::Tensor w, torch::Tensor b,
+                                           int64_t P) {
+    CHECK_IN(img); CHECK_IN(w); CHECK_IN(b);
+    const int64_t B = img.size(0), C = img.size(1), H = img.size(2),
+                  W = img.size(3);
+    const int64_t S = W / P, N = (H / P) * S;
+    const int64_t K0 = P * P * C, Kp = pad_k(K0), dout = w.size(0);
+    TORCH_CHECK(w.size(1) == K0, "embed weight/patch mismatch");
+    auto opts = img.options();
+    hipStream_t s = cur_stream();
+
+    auto X = torch::empty({B, N, Kp}, opts);
+    launch_patchify(img.data_ptr(), X.data_ptr(), (int)B, (int)C, (int)H,
+                    (int)W, (int)P, (int)Kp, s);
+    check_launch();
+    auto Wp = torch::empty({dout, Kp}, opts);
+    launch_pad_cols(w.data_ptr(), Wp.data_ptr(), dout, (int)K0, (int)Kp, s);
+    check_launch();
+
+    auto tokens = torch::empty({B, N, dout}, opts);
+    GemmParams p = base_params(B * N, dout, Kp, LAYOUT_NT, 1, 1, 1.0f);
+    p.A.base = X.data_ptr(); p.A.sin = 0; p.A.ld = Kp;
+    p.B.base = Wp.data_ptr(); p.B.sin = 0; p.B.ld = Kp;
+    p.Cbase = tokens.data_ptr(); p.Csin = 0; p.Cld = dout;
+    p.bias_base = b.data_ptr(); p.bias_sin = 0; p.has_bias = 1;
+    run_gemm(p, s, opts, true);
+    return {tokens, X};
+}
+
+std::vector<torch::Tensor> patch_embed_bwd(torch::Tensor dTokens,
+                                           torch::Tensor X, torch::Tensor w,
+                                           int64_t P, int64_t imgH,
+                                           int64_t imgW, bool need_dimg) {
+    CHECK_IN(dTokens); CHECK_IN(X); CHECK_IN(w);
+    const int64_t B = dTokens.size(0), N = dTokens.size(1),
+                  dout = dTokens.size(2);
+    const int64_t Kp = X.size(2), K0 = w.size(1), C = K0 / (P * P);
+    const int64_t M = B * N;
+    auto opts = dTokens.options();
+    hipStream_t s = cur_stream();
+
+    // dWp[o, k] = sum_m dTokens[m, o] * X[m, k]   (TN, split-K eligible)
+    auto dWp = torch::empty({dout, Kp}, opts);
+    {
+        GemmParams p = base_params(dout, Kp, M, LAYOUT_TN, 1, 1, 1.0f);
+        p.A.base = dTokens.data_ptr(); p.A.sin = 0; p.A.ld = dout;
+        p.B.base = X.data_ptr(); p.B.sin = 0; p.B.ld = Kp;
+        p.Cbase = dWp.data_ptr(); p.Csin = 0; p.Cld = Kp;
+        run_gemm(p, s, opts, true);
+    }
+    auto dW = torch::empty({dout, K0}, opts);
+    launch_slice_cols(dWp.data_ptr(), dW.data_ptr(), dout, (int)Kp,
+                      (int)K0, s);
+    check_launch();
+
+    auto dB = torch::empty({dout}, opts);
+    {
+        auto cws = torch::empty({(long)colsum_rb(M) * dout},
+                                opts.dtype(at::kFloat));
+        launch_colsum(dTokens.data_ptr(), cws.data_ptr<float>(),
+                      dB.data_ptr(), 1, M, dout, s);
+        check_launch();
+    }
+
+    torch::Tensor dImg = torch::empty({0}, opts);
+    if (need_dimg) {
+        auto Wp = torch::empty({dout, Kp}, opts);
+        launch_pad_cols(w.data_ptr(), Wp.data_ptr(), dout, (int)K0,
+                        (int)Kp, s);
+        check_launch();
+        // dX[m, k] = sum_o dTokens[m, o] * Wp[o, k]   (NN)
+        auto dX = torch::empty({B, N, Kp}, opts);
+        GemmParams p = base_params(M, Kp, dout, LAYOUT_NN, 1, 1, 1.0f);
+        p.A.base = dTokens.data_ptr(); p.A.sin = 0; p.A.ld = dout;
+        p.B.base = Wp.data_ptr(); p.B.sin = 0; p.B.ld = Kp;
+        p.Cbase = dX.data_ptr(); p.Csin = 0; p.Cld = Kp;
+        run_gemm(p, s, opts, true);
+        dImg = torch::empty({B, C, imgH, imgW}, opts);
+        launch_unpatchify(dX.data_ptr(), dImg.data_ptr(), (int)B, (int)C,
+                          (int)imgH, (int)imgW, (int)P, (int)Kp, s);
+        check_launch();
+    }
+    return {dW, dB, dImg};
+}
+
+torch::Tensor dpos(torch::Tensor dLevels) {
+    CHECK_IN(dLevels);
+    const int64_t B = dLevels.size(0), N = dLevels.size(1),
+                  L = dLevels.size(2), d = dLevels.size(3);
+    auto out = torch::empty({N, d}, dLevels.options());
+    launch_dpos(dLevels.data_ptr(), out.data_ptr(), (int)B, (int)N, (int)L,
+                (int)d, cur_stream());
+    check_launch();
+    return out;
+}
+
+// ------------------------------------------------------------------ //
+// fused AdamW: one kernel pass over all parameter tensors (bf16 grads ->
+// fp32 m/v/master -> bf16 param write-back) with the deterministic
+// global-norm grad clip fused. step_dev is a persistent device counter
+// (incremented on-device so the op is hipGraph-replay-safe).
+
+void fused_adamw(std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> masters,
+                 std::vector<torch::Tensor> m1s,
+                 std::vector<torch::Tensor> m2s,
+                 std::vector<torch::Tensor> params,
+                 double lr, double b1, double b2, double eps, double wd,
+                 double max_norm, torch::Tensor partials, torch::Tensor norm,
+                 torch::Tensor step_dev) {
+    const size_t nt = grads.size();
+    TORCH_CHECK(nt > 0 && nt <= OPT_MAX_T, "fused_adamw: 1..", OPT_MAX_T,
+                " tensors supported, got ", nt);
+    TORCH_CHECK(masters.size() == nt && m1s.size() == nt
+                && m2s.size() == nt && params.size() == nt);
+    TORCH_CHECK(partials.numel() >= OPT_NPART
+                && partials.scalar_type() == at::kFloat);
+    TORCH_CHECK(norm.numel() >= 1 && norm.scalar_type() == at::kFloat);
+    TORCH_CHECK(step_dev.numel() >= 1
+                && step_dev.scalar_type() == at::kFloat);
+    OptTable t;
+    std::memset(&t, 0, sizeof(t));
+    t.nt = (int)nt;
+    long cum = 0;
+    for (size_t i = 0; i < nt; i++) {
+        CHECK_IN(grads[i]); CHECK_IN(params[i]);
+        TORCH_CHECK(masters[i].is_contiguous()
+                    && masters[i].scalar_type() == at::kFloat);
+        TORCH_CHECK(m1s[i].is_contiguous() && m2s[i].is_contiguous());
+        const long n = grads[i].numel();
+        TORCH_CHECK(masters[i].numel() == n && m1s[i].numel() == n
+                    && m2s[i].numel() == n && params[i].numel() == n);
+        t.g[i] = (const unsigned short*)grads[i].data_ptr();
+        t.mw[i] = masters[i].data_ptr<float>();
+        t.m1[i] = m1s[i].data_ptr<float>();
+        t.m2[i] = m2s[i].data_ptr<float>();
+        t.pw[i] = (unsigned short*)params[i].data_ptr();
+        t.cum[i] = cum;
+        cum += n;
+    }
+    t.cum[nt] = cum;
+    hipStream_t s = cur_stream();
+    launch_grad_norm(t, partials.data_ptr<float>(), norm.data_ptr<float>(),
+                     step_dev.data_ptr<float>(), s);
+    check_launch();
+    launch_adamw(t, (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                 (float)max_norm, norm.data_ptr<float>(),
+                 step_dev.data_ptr<float>(), s);
+    check_launch();
 }
 
 std::string build_info() {
@@ -782,10 +1023,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("w2t") = c10::nullopt);
     m.def("consensus_fwd", &consensus_fwd, "consensus attention forward");
     m.def("consensus_bwd", &consensus_bwd, "consensus attention backward");
-    m.def("level_mix_fwd", &level_mix_fwd, "level mix forward");
+    m.def("level_mix_fwd", &level_mix_fwd, "level mix forward",
+          py::arg("levels"), py::arg("bu"), py::arg("td"), py::arg("cons"),
+          py::arg("slab") = c10::nullopt, py::arg("slab_idx") = 0);
     m.def("level_mix_bwd", &level_mix_bwd, "level mix backward");
-    m.def("glom_step_fwd", &glom_step_fwd, "full GLOM iteration forward");
+    m.def("glom_step_fwd", &glom_step_fwd, "full GLOM iteration forward",
+          py::arg("tokens"), py::arg("levels"), py::arg("pos"),
+          py::arg("bw1"), py::arg("bb1"), py::arg("bw2"), py::arg("bb2"),
+          py::arg("tw1"), py::arg("tb1"), py::arg("tw2"), py::arg("tb2"),
+          py::arg("attend_self"), py::arg("mask"),
+          py::arg("slab") = c10::nullopt, py::arg("slab_idx") = 0);
     m.def("glom_step_bwd", &glom_step_bwd, "full GLOM iteration backward");
+    m.def("patch_embed_fwd", &patch_embed_fwd,
+          "patchify + embed GEMM (K1)");
+    m.def("patch_embed_bwd", &patch_embed_bwd, "patch embed backward");
+    m.def("dpos", &dpos, "positional embedding gradient reduction");
+    m.def("fused_adamw", &fused_adamw,
+          "fused AdamW + global-norm grad clip (torch semantics)");
     m.def("add4_into", &add4_into, "fused 4-way elementwise sum");
     m.def("ff_bwd_dh", &ff_bwd_dh);
     m.def("ff_bwd_dx", &ff_bwd_dx);

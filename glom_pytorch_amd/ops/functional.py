@@ -16,6 +16,50 @@ import torch
 from glom_pytorch_amd.ops import _load_extension
 
 
+class PatchEmbedFn(torch.autograd.Function):
+    """Patchify + embed GEMM (K1; reference glom_pytorch.py:94-97,114):
+    the rearrange runs as a HIP gather kernel into a K-padded (B,N,Kp)
+    token matrix and the 588->dim projection on the tuned NT MFMA kernel.
+    Backward: split-K TN weight grad, native colsum bias grad, and (only
+    when the image itself needs grad) an NN GEMM + scatter back to
+    (B,3,H,W)."""
+
+    @staticmethod
+    def forward(ctx, img, w, b, patch_size):
+        ext = _load_extension()
+        tokens, X = ext.patch_embed_fwd(img.contiguous(), w, b, patch_size)
+        ctx.save_for_backward(X, w)
+        ctx.patch = patch_size
+        ctx.img_hw = (img.shape[2], img.shape[3])
+        return tokens
+
+    @staticmethod
+    def backward(ctx, dTokens):
+        ext = _load_extension()
+        X, w = ctx.saved_tensors
+        need_dimg = ctx.needs_input_grad[0]
+        dW, dB, dImg = ext.patch_embed_bwd(
+            dTokens.contiguous(), X, w, ctx.patch, ctx.img_hw[0],
+            ctx.img_hw[1], need_dimg)
+        return (dImg if need_dimg else None), dW, dB, None
+
+
+class TrajectoryFn(torch.autograd.Function):
+    """return_all trajectory (reference glom_pytorch.py:126,145-148)
+    without the torch.stack copy: every step already wrote its output into
+    slab[t+1] (and slab[0] holds the initial state), so forward is an
+    identity on the slab and backward just slices the incoming gradient
+    back to the per-step outputs."""
+
+    @staticmethod
+    def forward(ctx, slab, *steps):
+        return slab
+
+    @staticmethod
+    def backward(ctx, dtraj):
+        return (None,) + tuple(dtraj[t] for t in range(dtraj.shape[0]))
+
+
 class GroupedFFFn(torch.autograd.Function):
     """Grouped per-level MLP (d -> 4d -> GELU -> d), bottom-up or top-down.
 
@@ -44,8 +88,9 @@ class GroupedFFFn(torch.autograd.Function):
         dPos = None
         if ctx.mode == 1 and pos is not None and ctx.needs_input_grad[2]:
             # pos was added to every top-down group input; its grad is the
-            # sum of the level-slice grads over batch and groups.
-            dPos = dLevels[:, :, 1:, :].sum(dim=(0, 2))
+            # sum of the level-slice grads over batch and groups (native
+            # deterministic reduction).
+            dPos = ext.dpos(dLevels)
         if ctx.mode == 0:
             return dTokens, dLevels, None, dW1, dB1, dW2, dB2, None
         return None, dLevels, dPos, dW1, dB1, dW2, dB2, None
@@ -108,13 +153,16 @@ class GlomStepFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, tokens, levels, pos, bw1, bb1, bw2, bb2,
                 tw1, tb1, tw2, tb2, attend_self, mask,
-                bw1t, bw2t, tw1t, tw2t):
+                bw1t, bw2t, tw1t, tw2t, slab_ref=None):
+        # slab_ref: optional (slab, idx) — write the step output straight
+        # into the preallocated trajectory slab (untracked alias return)
         ext = _load_extension()
+        slab, sidx = slab_ref if slab_ref is not None else (None, 0)
         if (torch.is_grad_enabled()
                 and os.environ.get("GLOM_FWD_STREAMS", "0") != "1"):
             out, bhp, bha, thp, tha, probs, rnorm = ext.glom_step_fwd(
                 tokens, levels, pos, bw1, bb1, bw2, bb2, tw1, tb1, tw2,
-                tb2, attend_self, mask)
+                tb2, attend_self, mask, slab, sidx)
         else:
             # inference: the three per-iteration chains (bottom-up MLP,
             # top-down MLP, consensus attention) only depend on `levels`;
@@ -141,7 +189,7 @@ class GlomStepFn(torch.autograd.Function):
             # blocks until the main stream catches up (allocator safety)
             for t in (tdY, thp, tha, cons, probs, rnorm):
                 t.record_stream(cur)
-            out = ext.level_mix_fwd(levels, buY, tdY, cons)
+            out = ext.level_mix_fwd(levels, buY, tdY, cons, slab, sidx)
         ctx.save_for_backward(tokens, levels, pos, bw1, bw2, tw1, tw2,
                               bhp, bha, thp, tha, probs, rnorm, mask,
                               bw1t, bw2t, tw1t, tw2t)
@@ -161,7 +209,7 @@ class GlomStepFn(torch.autograd.Function):
                 bw1t, bw2t, tw1t, tw2t)
             return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
                     dtw1, dtb1, dtw2, dtb2, None, None, None, None, None,
-                    None)
+                    None, None)
         if os.environ.get("GLOM_BWD_FORK", "3") == "3":
             cur = torch.cuda.current_stream()
             s_td, s_at, _ = GlomStepFn._side_streams()
@@ -187,10 +235,10 @@ class GlomStepFn(torch.autograd.Function):
                     t.record_stream(cur)
             dLevels = torch.empty_like(levels)
             ext.add4_into(dmix, bu[1], td[1], dAttn, dLevels)
-            dPos = td[1][:, :, 1:, :].sum(dim=(0, 2))
+            dPos = ext.dpos(td[1])
             return (bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
                     td[2], td[3], td[4], td[5], None, None, None, None,
-                    None, None)
+                    None, None, None)
         # 4-way fork variant (weight grads on a dedicated stream): measured
         # slightly SLOWER than the 3-way fork above (sync overhead), kept
         # behind GLOM_BWD_FORK=4 for future tuning.
@@ -234,10 +282,10 @@ class GlomStepFn(torch.autograd.Function):
             t.record_stream(cur)
         dLevels = torch.empty_like(levels)
         ext.add4_into(dmix, bu_dl, td_dl, dAttn, dLevels)
-        dPos = td_dl[:, :, 1:, :].sum(dim=(0, 2))
+        dPos = ext.dpos(td_dl)
         return (bu_dt, dLevels, dPos, bu_w1, bu_db1, bu_w2, bu_b2,
                 td_w1, td_db1, td_w2, td_b2, None, None, None, None, None,
-                None)
+                None, None)
 
 
 def _transposed_weights(model):
@@ -258,7 +306,7 @@ def _transposed_weights(model):
         )
 
 
-def glom_step(model, tokens, levels, pos, mask, wts=None):
+def glom_step(model, tokens, levels, pos, mask, wts=None, slab_ref=None):
     bw = model.bottom_up.net
     tw = model.top_down.net
     if wts is None:
@@ -267,14 +315,17 @@ def glom_step(model, tokens, levels, pos, mask, wts=None):
         tokens, levels, pos,
         bw[1].weight[..., 0], bw[1].bias, bw[3].weight[..., 0], bw[3].bias,
         tw[1].weight[..., 0], tw[1].bias, tw[3].weight[..., 0], tw[3].bias,
-        model.attention.attend_self, mask, *wts)
+        model.attention.attend_self, mask, *wts, slab_ref)
 
 
 def glom_forward(model, img, iters, levels=None, return_all=False):
     from glom_pytorch_amd.utils.profiling import trace_range
     b = img.shape[0]
     with trace_range("glom/patch_embed"):
-        tokens = model.image_to_tokens(img)      # K1: once per forward
+        # K1: hand-written patchify + NT MFMA GEMM, once per forward
+        emb = model.image_to_tokens[1]
+        tokens = PatchEmbedFn.apply(img, emb.weight, emb.bias,
+                                    model.patch_size)
     n = tokens.shape[1]
     pos = model.pos_emb.weight
     mask = (model.attention.non_local_mask
@@ -288,13 +339,25 @@ def glom_forward(model, img, iters, levels=None, return_all=False):
 
     wts = _transposed_weights(model) if torch.is_grad_enabled() else None
 
-    trajectory = [levels]
+    # return_all: steps write straight into a preallocated (T+1) slab —
+    # no torch.stack copy at the end (SURVEY.md §2.3 note)
+    slab = None
+    if return_all:
+        slab = torch.empty((iters + 1,) + tuple(levels.shape),
+                           device=levels.device, dtype=levels.dtype)
+        with torch.no_grad():
+            slab[0].copy_(levels)
+
+    steps = [levels]
     with trace_range(f"glom/iterate x{iters}"):
         for t in range(iters):
             with trace_range(f"glom/step{t}"):
-                levels = glom_step(model, tokens, levels, pos, mask, wts)
-            trajectory.append(levels)
+                levels = glom_step(
+                    model, tokens, levels, pos, mask, wts,
+                    (slab, t + 1) if slab is not None else None)
+            if return_all:
+                steps.append(levels)
 
     if return_all:
-        return torch.stack(trajectory)
+        return TrajectoryFn.apply(slab, *steps)
     return levels

@@ -2,13 +2,16 @@
 recipe, README.md:56-90): noise the image, run the model with return_all,
 decode the top level at a mid-iteration timestep back to pixels, MSE against
 the clean image. Adds what the reference leaves to the user: optimizer,
-data-parallel gradient sync, checkpointing, metrics."""
+data-parallel gradient sync, checkpointing, metrics — and, on the native
+GPU path, hipGraph capture of the ENTIRE training step (noise + forward +
+backward + fused AdamW), replayed as a single graph launch per step."""
 
 from __future__ import annotations
 
 import json
 import os
 import time
+import warnings
 
 import torch
 import torch.distributed as dist
@@ -35,12 +38,17 @@ class DenoisingDecoder(nn.Sequential):
 class DenoisingTrainer:
     def __init__(self, model, *, lr=3e-4, noise_std=1.0, decode_step=7,
                  grad_clip=1.0, distributed=False, bucket_bytes=16 << 20,
-                 process_group=None, log_path=None):
+                 process_group=None, graph_step=True, log_path=None):
         """``process_group``: the group whose ranks are data-parallel
         replicas (default: the global group). For 2-D DP x SP meshes pass
         the DP subgroup here — BucketedDDP AVERAGES over its group, which
         is correct for DP replicas; SP shards need a separate SUM
-        all-reduce over the SP subgroup (see parallel/sequence.py)."""
+        all-reduce over the SP subgroup (see parallel/sequence.py).
+
+        ``graph_step``: hipGraph-capture the whole training step on the
+        native bf16 GPU path (keyed on batch shape + iters) and replay it;
+        falls back to eager launches automatically if capture fails.
+        Disable with graph_step=False or GLOM_NO_GRAPH_STEP=1."""
         self.model = model
         self.dim = model.dim
         self.decode_step = decode_step
@@ -59,19 +67,41 @@ class DenoisingTrainer:
             # rank-liveness probe: a hung peer becomes a clean abort
             self.heartbeat = Heartbeat(every_steps=50)
         params = list(model.parameters()) + list(self.decoder.parameters())
+        self._params = params
         # bf16 training keeps fp32 master weights: updates of size lr*grad
         # would otherwise partially round away in bf16 parameter storage
         self.master = None
+        self.fused_opt = None
         if p.dtype == torch.bfloat16:
             self.master = [q.detach().float().requires_grad_(False)
                            for q in params]
-            self._params = params
-            self.opt = torch.optim.AdamW(self.master, lr=lr, foreach=True)
+            if p.is_cuda and self._ops_available():
+                # single-kernel AdamW: bf16 grads -> fp32 m/v/master ->
+                # bf16 params, with the global-norm clip fused
+                from glom_pytorch_amd.ops.optim import FusedAdamW
+                self.fused_opt = FusedAdamW(
+                    params, self.master, lr=lr, weight_decay=1e-2,
+                    max_grad_norm=grad_clip if grad_clip else 0.0)
+                self.opt = self.fused_opt
+            else:
+                self.opt = torch.optim.AdamW(self.master, lr=lr,
+                                             foreach=True)
         else:
             self.opt = torch.optim.AdamW(params, lr=lr, foreach=True)
         self.grad_clip = grad_clip
         self.step_idx = 0
         self.log_path = log_path
+        self.graph_step = (graph_step
+                           and os.environ.get("GLOM_NO_GRAPH_STEP", "0")
+                           != "1")
+        self._graphs: dict = {}
+
+    @staticmethod
+    def _ops_available():
+        from glom_pytorch_amd import ops
+        return ops.available()
+
+    # ------------------------------ stepping -------------------------- #
 
     def step(self, img: torch.Tensor, iters: int | None = None,
              sync_loss: bool = True):
@@ -79,6 +109,22 @@ class DenoisingTrainer:
         returned as a device tensor (no host sync), letting the host run
         ahead and queue the next step's launches."""
         iters = iters if iters is not None else 2 * self.model.levels
+        if self._graph_usable(img):
+            loss = self._graphed_step(img, iters)
+        else:
+            loss = self._eager_step(img, iters)
+        self.step_idx += 1
+        if self.heartbeat is not None:
+            self.heartbeat.tick()
+        return loss.item() if sync_loss else loss.detach()
+
+    def _graph_usable(self, img) -> bool:
+        return (self.graph_step and self.fused_opt is not None
+                and img.is_cuda and img.dtype == torch.bfloat16
+                and not getattr(self.model, "force_eager", False)
+                and os.environ.get("GLOM_FORCE_EAGER", "0") != "1")
+
+    def _eager_step(self, img: torch.Tensor, iters: int) -> torch.Tensor:
         t = min(self.decode_step, iters)
         self.opt.zero_grad(set_to_none=True)
         noised = img + torch.randn_like(img) * self.noise_std
@@ -90,8 +136,9 @@ class DenoisingTrainer:
         if self.distributed:
             self.ddp_model.finalize()
             self.ddp_dec.finalize()
-            self.heartbeat.tick()
-        if self.master is not None:
+        if self.fused_opt is not None:
+            self.fused_opt.step()
+        elif self.master is not None:
             with torch.no_grad():
                 for mw, q in zip(self.master, self._params):
                     # None (not stale) when the param got no grad this
@@ -109,8 +156,70 @@ class DenoisingTrainer:
                     [q for g in self.opt.param_groups for q in g["params"]],
                     self.grad_clip)
             self.opt.step()
-        self.step_idx += 1
-        return loss.item() if sync_loss else loss.detach()
+        return loss.detach()
+
+    # --------------------- hipGraph-captured step --------------------- #
+
+    def _graphed_step(self, img: torch.Tensor, iters: int) -> torch.Tensor:
+        key = (tuple(img.shape), iters)
+        entry = self._graphs.get(key)
+        if entry is None:
+            entry = self._capture(img, iters)
+            self._graphs[key] = entry
+        if entry is False:
+            return self._eager_step(img, iters)
+        entry["img"].copy_(img)
+        entry["graph"].replay()
+        self.fused_opt.note_replays(1)
+        return entry["loss"]
+
+    def _capture(self, img: torch.Tensor, iters: int):
+        """Record noise + forward (all `iters` GLOM steps) + backward +
+        bucketed all-reduce (if distributed) + fused AdamW as ONE hipGraph.
+        torch.cuda.graph handles the philox RNG state, so every replay
+        draws fresh noise. Any capture failure falls back to eager."""
+        if self.distributed and os.environ.get("GLOM_GRAPH_DIST",
+                                               "1") == "0":
+            return False
+        try:
+            static_img = img.clone()
+            # snapshot optimizer/weight state: the warmup steps below are
+            # real updates, rolled back so a graphed step == one step
+            fo = self.fused_opt
+            with torch.no_grad():
+                snap = ([m.clone() for m in self.master]
+                        + [t.clone() for t in fo.exp_avg]
+                        + [t.clone() for t in fo.exp_avg_sq]
+                        + [q.detach().clone() for q in self._params]
+                        + [fo._step_dev.clone()])
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):     # allocator/runtime warmup
+                    self._eager_step(static_img, iters)
+            torch.cuda.current_stream().wait_stream(s)
+            with torch.no_grad():
+                live = (self.master + fo.exp_avg + fo.exp_avg_sq
+                        + [q.data for q in self._params] + [fo._step_dev])
+                for dst, src in zip(live, snap):
+                    dst.copy_(src)
+                fo._step_host = max(0, fo._step_host - 2)
+            # start from grads=None: backward inside capture then ASSIGNS
+            # fresh graph-pool grad tensors (no accumulate node recorded)
+            for q in self._params:
+                q.grad = None
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                loss = self._eager_step(static_img, iters)
+            # capture RECORDS the step without executing it: undo the
+            # Python-side count bump from the captured call
+            fo._step_host = max(0, fo._step_host - 1)
+            return {"graph": graph, "img": static_img, "loss": loss}
+        except Exception as e:  # pragma: no cover - GPU/runtime specific
+            warnings.warn(f"training-step hipGraph capture failed "
+                          f"({type(e).__name__}: {e}); using eager "
+                          f"launches")
+            return False
 
     def log(self, **metrics):
         if self.log_path:
@@ -145,7 +254,15 @@ class DenoisingTrainer:
             for mw, saved in zip(self.master, ckpt["master"]):
                 mw.data.copy_(saved.to(mw.device))
         self.step_idx = ckpt["step"]
+        # params follow the restored masters (the fused path reads/writes
+        # masters as the source of truth)
+        if self.master is not None:
+            with torch.no_grad():
+                for mw, q in zip(self.master, self._params):
+                    q.data.copy_(mw.to(q.dtype))
         torch.set_rng_state(ckpt["rng"])
         if ckpt.get("cuda_rng") is not None and torch.cuda.is_available():
             torch.cuda.set_rng_state(ckpt["cuda_rng"])
+        # drop any captured graphs: they bake in old grad/param pointers
+        self._graphs.clear()
         return self

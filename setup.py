@@ -20,6 +20,7 @@ ext = CUDAExtension(
         "glom_pytorch_amd/ops/csrc/gemm.hip",
         "glom_pytorch_amd/ops/csrc/gemm_fast.hip",
         "glom_pytorch_amd/ops/csrc/aux_kernels.hip",
+        "glom_pytorch_amd/ops/csrc/native_ops.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
