@@ -42,6 +42,19 @@ hipStream_t cur_stream() {
     return c10::hip::getCurrentHIPStream().stream();
 }
 
+// nt8p dispatch mode: opt-in (GLOM_NT8P=1), runtime-settable for
+// within-process A/B microbenches (guide §5.4 rule 24). Measured: +15%
+// on the down-projection microbench but -1% END-TO-END — the 128 KiB
+// LDS / 1-block-per-CU tile starves the concurrently running backward
+// streams of co-resident blocks, so the step's 2-stream overlap loses
+// more than the kernel gains (profiles/README.md round-2 notes).
+int g_nt8p_mode = []() {
+    const char* e = getenv("GLOM_NT8P");
+    return (e && e[0] == '1') ? 1 : 0;
+}();
+
+void set_nt8p(bool on) { g_nt8p_mode = on ? 1 : 0; }
+
 void check_launch() {
     hipError_t e = hipGetLastError();
     TORCH_CHECK(e == hipSuccess, "HIP launch failed: ", hipGetErrorString(e));
@@ -100,14 +113,31 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
         const char* e = getenv("GLOM_DISPATCH_DEBUG");
         return e && e[0] == '1';
     }();
+    // 8-phase 256^2 schedule (guide §5 template): per-phase interleave of
+    // ds_read / glds / MFMA with counted per-wave vmcnt. GLOM_NT8P=0
+    // opts out; set_nt8p() lets microbenches A/B within one process.
+    const bool nt8p_on = g_nt8p_mode == 1;
+    // measured (profiles/README.md): the 256^2 8-phase tile wins at
+    // K >= 2048 (down-projection 685 -> 789 TF, square parity+), while
+    // the K=512 up-projection family stays on nt5p's persistent ring
+    // (8-phase loses its prologue amortization at 8 K-steps/tile)
+    const bool nt8p = nt8p_on && p.layout == LAYOUT_NT && lds_ok && no_xform
+                      && p.M % 256 == 0 && p.N % 256 == 0 && p.K % 64 == 0
+                      && p.K >= 2048
+                      && !(p.Cflags & OP_TABLE)
+                      && (p.epilogue == EPI_NONE
+                          || p.epilogue == EPI_GELUGRAD)
+                      && (long)(p.N / 256) * (p.M / 256) * p.nproblems >= 256;
     if (disp_dbg)
         fprintf(stderr,
                 "[dispatch] L%d M%ld N%ld K%ld epi%d nt3=%d nt5p=%d on=%d "
-                "ctabflag=%d m512=%d\n",
+                "nt8p=%d ctabflag=%d m512=%d\n",
                 p.layout, (long)p.M, (long)p.N, (long)p.K, p.epilogue,
-                (int)nt3, (int)nt5p, (int)nt5p_on, (int)(p.Cflags & OP_TABLE),
-                (int)(p.M % 512 == 0));
-    if (nt5p)
+                (int)nt3, (int)nt5p, (int)nt5p_on, (int)nt8p,
+                (int)(p.Cflags & OP_TABLE), (int)(p.M % 512 == 0));
+    if (nt8p)
+        launch_gemm_nt_fast8p(p, s);
+    else if (nt5p)
         launch_gemm_nt_fast5p(p, s);
     else if (nt3)
         launch_gemm_nt_fast4(p, s);   // 3-ring counted-vmcnt variant
@@ -1044,6 +1074,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ff_bwd_dh", &ff_bwd_dh);
     m.def("ff_bwd_dx", &ff_bwd_dx);
     m.def("ff_bwd_dw", &ff_bwd_dw);
+    m.def("set_nt8p", &set_nt8p, "toggle the 8-phase NT kernel (A/B)");
     m.def("build_info", &build_info);
     m.def("bench_gemm", &bench_gemm, "raw GEMM microbench (tuning only)");
 }

@@ -82,4 +82,5 @@ void launch_gemm_nn_fast(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast3(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast4(const GemmParams& p, hipStream_t stream);
 void launch_gemm_nt_fast5p(const GemmParams& p, hipStream_t stream);
+void launch_gemm_nt_fast8p(const GemmParams& p, hipStream_t stream);
 void launch_gemm_tn_sk(const GemmParams& p, hipStream_t stream);

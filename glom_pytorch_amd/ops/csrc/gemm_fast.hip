@@ -1415,6 +1415,243 @@ void launch_gemm_nt_fast5p(const GemmParams& p, hipStream_t stream) {
 
 
 
+// --------------------------------------------------------------------- //
+// gemm_nt_fast8p: 256x256x64 tile, 8 waves (2M x 4N, 128x64 per wave),
+// guide §5 "256² 8-phase template" schedule: the K-step is split into 4
+// phases of {ds_read subtile || 2 global_load_lds -> barrier ->
+// setprio(1) 16xMFMA setprio(0) -> counted per-wave vmcnt -> barrier}.
+// 2 full-tile LDS buffers (128 KiB, 1 block/CU); each wave stages ONLY
+// the data it will itself read (its B quarter + the A mfrag-pair of its
+// wn slot), ordered B-first, so the counted waits leave later-needed
+// halves in flight across phase barriers (T3+T4) and the phase split
+// gives setprio something to arbitrate (T5).
+//
+// Per-wave wait schedule (steady state; derivation in-line):
+//   end of phase 0: wave wn==1 waits vmcnt(2)   (its A(t) share)
+//   end of phase 1: wave wn==2 waits vmcnt(4)
+//   end of phase 2: wave wn==3 waits vmcnt(6)
+//   end of phase 3: wave wn==0 waits vmcnt(0), others vmcnt(4)
+// When the next tile is not staged (last K-tile) the protecting counts
+// drop to 0/0/0/0 (nothing newer in the queue to leave in flight).
+
+#define BM8 256
+#define BN8 256
+#define NT8 512
+
+__global__ __launch_bounds__(NT8) void gemm_nt_fast8p_kernel(GemmParams p) {
+    __shared__ ushort_t smem[2 * (BM8 + BN8) * FBK];   // 128 KiB
+    const int abuf = BM8 * FBK, stride = (BM8 + BN8) * FBK;
+
+    const int pid = blockIdx.z;
+    int nwg = gridDim.x * gridDim.y;
+    int bid = blockIdx.y * gridDim.x + blockIdx.x;
+    {
+        int q = nwg >> 3, r = nwg & 7, xcd = bid & 7, off = bid >> 3;
+        bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+    }
+    const int n0 = (bid / gridDim.y) * BN8;   // column-major: n outer
+    const int m0 = (bid % gridDim.y) * BM8;
+
+    const ushort_t* Ap;
+    const ushort_t* Bp;
+    long lda, ldb;
+    resolve_ptr2(p.A, p.Atab, p.Atabld, pid, p.nInner, &Ap, &lda);
+    resolve_ptr2(p.B, p.Btab, p.Btabld, pid, p.nInner, &Bp, &ldb);
+
+    ushort_t* Cp;
+    long ldc;
+    {
+        const ushort_t* tmp;
+        OpArg ca;
+        ca.base = p.Cbase; ca.sin = p.Csin; ca.sout = p.Csout; ca.ld = p.Cld;
+        ca.flags = p.Cflags;
+        resolve_ptr2(ca, (const void* const*)p.Ctab, p.Ctabld, pid, p.nInner,
+                     &tmp, &ldc);
+        Cp = (ushort_t*)tmp;
+    }
+    const ushort_t* biasp = nullptr;
+    if (p.has_bias)
+        biasp = (const ushort_t*)p.bias_base
+                + (long)(pid % p.nInner) * p.bias_sin
+                + (long)(pid / p.nInner) * p.bias_sout;
+    const float* csp = nullptr;
+    if (p.has_colscale)
+        csp = (const float*)p.colscale_base
+              + (long)(pid % p.nInner) * p.cs_sin
+              + (long)(pid / p.nInner) * p.cs_sout;
+    const ushort_t* auxp = nullptr;
+    if (p.epilogue == EPI_GELUGRAD)
+        auxp = (const ushort_t*)p.aux_base
+               + (long)(pid % p.nInner) * p.aux_sin
+               + (long)(pid / p.nInner) * p.aux_sout;
+    float* colp = nullptr;
+    if (p.colsum_out)
+        colp = p.colsum_out + (long)(pid % p.nInner) * p.colsum_sin;
+
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int whalf = wid >> 2;          // A half (0: rows 0-127, 1: 128-255)
+    const int wq = wid & 3;              // B quarter / A stage slot
+    const int wm = whalf * 128;
+    const int wn = wq * 64;
+    const int lrow = lane & 15;
+    const int kq = lane >> 4;
+
+    float csv[4] = {1.f, 1.f, 1.f, 1.f};
+    float bvv[4] = {0.f, 0.f, 0.f, 0.f};
+    if (csp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            csv[j16] = csp[n0 + wn + j16 * 16 + lrow];
+    }
+    if (biasp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            bvv[j16] = bf2f(biasp[n0 + wn + j16 * 16 + lrow]);
+    }
+
+    f32x4 acc[8][4] = {};
+
+    // Stage 2 chunks (16 rows, 2 glds/thread) of this wave's OWN data.
+    // part 0/1: B quarter rows [wq*64 + whalf*32 + 16*part)
+    // part 2/3: A mfrag-pair rows [whalf*128 + wq*32 + 16*(part-2))
+    auto stage2 = [&](int buf, int k0, int part) {
+        ushort_t* img;
+        int row0;
+        if (part < 2) {
+            img = smem + buf * stride + abuf;            // B image
+            row0 = wq * 64 + whalf * 32 + part * 16;
+        } else {
+            img = smem + buf * stride;                    // A image
+            row0 = whalf * 128 + wq * 32 + (part - 2) * 16;
+        }
+        const ushort_t* src = (part < 2 ? Bp : Ap);
+        const long ld = (part < 2 ? ldb : lda);
+        const int g0 = (part < 2 ? n0 : m0);
+#pragma unroll
+        for (int c = 0; c < 2; c++) {
+            int row = row0 + c * 8 + (lane >> 3);
+            int swz8 = ((lane & 7) ^ swz_row(row)) * 8;
+            const ushort_t* g = src + (long)(g0 + row) * ld + k0 + swz8;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)g,
+                (__attribute__((address_space(3))) unsigned int*)
+                    (img + (row >> 3) * 512), 16, 0, 0);
+        }
+    };
+
+    const int nk = p.K / FBK;
+    // prologue: stage tile 0 fully, drain, enter the loop (tile t's
+    // phases stage tile t+1 into the buffer freed at the end of t-1)
+#pragma unroll
+    for (int part = 0; part < 4; part++) stage2(0, 0, part);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (int t = 0; t < nk; t++) {
+        const ushort_t* Al = smem + (t & 1) * stride;
+        const ushort_t* Bl = Al + abuf;
+        const bool more = (t + 1) < nk;   // staging tile t+1 this pass
+        short8 bfrag[2][4];
+#pragma unroll
+        for (int s = 0; s < 2; s++)
+#pragma unroll
+            for (int j = 0; j < 4; j++) {
+                int row = wn + j * 16 + lrow;
+                int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                bfrag[s][j] = *(const short8*)&Bl[row * FBK + off];
+            }
+#pragma unroll
+        for (int ph = 0; ph < 4; ph++) {
+            short8 af[2][2];
+#pragma unroll
+            for (int s = 0; s < 2; s++)
+#pragma unroll
+                for (int i = 0; i < 2; i++) {
+                    int row = wm + (2 * ph + i) * 16 + lrow;
+                    int off = (s * 32 + kq * 8) ^ (swz_row(row) << 3);
+                    af[s][i] = *(const short8*)&Al[row * FBK + off];
+                }
+            if (more) stage2((t + 1) & 1, (t + 1) * FBK, ph);
+            __builtin_amdgcn_s_barrier();
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int s = 0; s < 2; s++)
+#pragma unroll
+                for (int i = 0; i < 2; i++)
+#pragma unroll
+                    for (int j = 0; j < 4; j++)
+                        acc[2 * ph + i][j] =
+                            __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                af[s][i], bfrag[s][j], acc[2 * ph + i][j],
+                                0, 0, 0);
+            __builtin_amdgcn_s_setprio(0);
+            // per-wave counted wait protecting the NEXT phase's ds_reads
+            if (ph < 3) {
+                if (__builtin_amdgcn_readfirstlane(wq) == ph + 1) {
+                    if (!more)
+                        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                    else if (ph == 0)
+                        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+                    else if (ph == 1)
+                        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+                    else
+                        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+                }
+            } else if (t + 1 < nk) {
+                // tile boundary: everyone needs B(t+1); wn==0 also its
+                // A(t+1) mfrag 0-1 share (the last 4 it issued)
+                if (__builtin_amdgcn_readfirstlane(wq) == 0)
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                else
+                    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+            }
+            __builtin_amdgcn_s_barrier();
+        }
+    }
+
+    // ------------- register epilogue (mirrors nt5p, 8 mfrags) ----------
+    float colacc[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int i16 = 0; i16 < 8; i16++) {
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+            long gi = m0 + wm + i16 * 16 + kq * 4 + r;
+            float vv[4];
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++)
+                vv[j16] = acc[i16][j16][r] * p.alpha * csv[j16];
+            if (auxp) {
+                const ushort_t* auxrow = auxp + gi * p.aux_ld;
+                float gx[4], gy[4];
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++)
+                    gx[j16] = bf2f(auxrow[n0 + wn + j16 * 16 + lrow]);
+                gelu_grad_vec<4>(gx, gy);
+#pragma unroll
+                for (int j16 = 0; j16 < 4; j16++) vv[j16] *= gy[j16];
+            }
+            ushort_t* crow = Cp + gi * ldc + n0 + wn + lrow;
+#pragma unroll
+            for (int j16 = 0; j16 < 4; j16++) {
+                ushort_t cb = f2bf(vv[j16] + bvv[j16]);
+                colacc[j16] += bf2f(cb);
+                crow[j16 * 16] = cb;
+            }
+        }
+    }
+    if (colp) {
+#pragma unroll
+        for (int j16 = 0; j16 < 4; j16++)
+            atomicAdd(&colp[n0 + wn + j16 * 16 + lrow], colacc[j16]);
+    }
+}
+
+void launch_gemm_nt_fast8p(const GemmParams& p, hipStream_t stream) {
+    dim3 grid(p.N / BN8, p.M / BM8, p.nproblems);
+    hipLaunchKernelGGL(gemm_nt_fast8p_kernel, grid, dim3(NT8), 0, stream, p);
+}
+
 // ------- split-K-only TN variant (32 KiB arena, 4 blocks/CU) -------
 __global__ __launch_bounds__(NTHREADS) void gemm_tn_sk_kernel(GemmParams p) {
     __shared__ ushort_t smem[2 * BM * FBK];   // 32 KiB: 4 blocks/CU
