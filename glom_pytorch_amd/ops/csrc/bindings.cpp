@@ -245,7 +245,10 @@ std::vector<torch::Tensor> grouped_ff_fwd(
         p.bias_base = b2.data_ptr(); p.bias_sin = d; p.has_bias = 1;
         run_gemm(p, s, opts, true);
     }
-    return {Y, Hpre, Hact};
+    if (mode != 1) td_in = torch::empty({0}, opts);
+    // td_in returned so the backward reuses it (saves re-running
+    // k_add_pos per iteration in the weight-grad GEMM)
+    return {Y, Hpre, Hact, td_in};
 }
 
 std::vector<torch::Tensor> grouped_ff_bwd(
@@ -254,7 +257,8 @@ std::vector<torch::Tensor> grouped_ff_bwd(
         torch::Tensor w1, torch::Tensor w2, torch::Tensor Hpre,
         torch::Tensor Hact, int64_t mode,
         c10::optional<torch::Tensor> w1t_opt = c10::nullopt,
-        c10::optional<torch::Tensor> w2t_opt = c10::nullopt) {
+        c10::optional<torch::Tensor> w2t_opt = c10::nullopt,
+        c10::optional<torch::Tensor> td_in_opt = c10::nullopt) {
     CHECK_IN(dY); CHECK_IN(levels); CHECK_IN(w1); CHECK_IN(w2); CHECK_IN(Hpre);
     CHECK_IN(Hact);
     const int64_t B = levels.size(0), N = levels.size(1),
@@ -286,12 +290,17 @@ std::vector<torch::Tensor> grouped_ff_bwd(
 
     torch::Tensor td_in;
     if (mode == 1) {
-        auto pos = pos_opt.value();
-        CHECK_IN(pos);
-        td_in = torch::empty({B, N, G, d}, opts);
-        launch_add_pos(levels.data_ptr(), pos.data_ptr(), td_in.data_ptr(),
-                       td_in.numel(), (int)N, (int)L, (int)d, s);
-        check_launch();
+        if (td_in_opt.has_value() && td_in_opt->numel() > 0) {
+            td_in = td_in_opt.value();   // saved by the forward
+        } else {
+            auto pos = pos_opt.value();
+            CHECK_IN(pos);
+            td_in = torch::empty({B, N, G, d}, opts);
+            launch_add_pos(levels.data_ptr(), pos.data_ptr(),
+                           td_in.data_ptr(), td_in.numel(), (int)N, (int)L,
+                           (int)d, s);
+            check_launch();
+        }
     }
 
     // dHpre = (dY_g @ W2_g) * gelu'(Hpre_g)  -- NT with W2^T; when the
@@ -641,7 +650,7 @@ std::vector<torch::Tensor> glom_step_fwd(
                              tb2, 1);
     auto at = consensus_fwd(levels, attend_self, mask);
     auto out = level_mix_fwd(levels, bu[0], td[0], at[0], slab, slab_idx);
-    return {out, bu[1], bu[2], td[1], td[2], at[1], at[2]};
+    return {out, bu[1], bu[2], td[1], td[2], at[1], at[2], td[3]};
 }
 
 std::vector<torch::Tensor> glom_step_bwd(
@@ -653,13 +662,14 @@ std::vector<torch::Tensor> glom_step_bwd(
         c10::optional<torch::Tensor> mask,
         c10::optional<torch::Tensor> bw1t, c10::optional<torch::Tensor> bw2t,
         c10::optional<torch::Tensor> tw1t,
-        c10::optional<torch::Tensor> tw2t) {
+        c10::optional<torch::Tensor> tw2t,
+        c10::optional<torch::Tensor> td_in = c10::nullopt) {
     const int64_t L = levels.size(2);
     auto mix = level_mix_bwd(dnew);        // {dmix, dtd}
     auto bu = grouped_ff_bwd(mix[0], tokens, levels, c10::nullopt, bw1,
                              bw2, buHpre, buHact, 0, bw1t, bw2t);
     auto td = grouped_ff_bwd(mix[1], c10::nullopt, levels, pos, tw1, tw2,
-                             tdHpre, tdHact, 1, tw1t, tw2t);
+                             tdHpre, tdHact, 1, tw1t, tw2t, td_in);
     auto dAttn = consensus_bwd(mix[0], levels, probs, rnorm, attend_self,
                                mask);
     auto dLevels = torch::empty_like(levels);
@@ -672,8 +682,10 @@ std::vector<torch::Tensor> glom_step_bwd(
     const int64_t B = levels.size(0), N = levels.size(1),
                   d = levels.size(3);
     auto dPos = torch::empty({N, d}, levels.options());
-    launch_dpos(td[1].data_ptr(), dPos.data_ptr(), (int)B, (int)N, (int)L,
-                (int)d, cur_stream());
+    const int BCH = (int)std::min<int64_t>(B, 8);
+    auto ws = torch::empty({BCH * N * d}, levels.options().dtype(at::kFloat));
+    launch_dpos(td[1].data_ptr(), dPos.data_ptr(), ws.data_ptr<float>(),
+                BCH, (int)B, (int)N, (int)L, (int)d, cur_stream());
     check_launch();
     return {bu[0], dLevels, dPos, bu[2], bu[3], bu[4], bu[5],
             td[2], td[3], td[4], td[5]};
@@ -822,7 +834,8 @@ std::vector<torch::Tensor> ff_bwd_dw(
         torch::Tensor dY, torch::Tensor dHpre,
         c10::optional<torch::Tensor> tokens_opt, torch::Tensor levels,
         c10::optional<torch::Tensor> pos_opt, torch::Tensor Hact,
-        int64_t mode) {
+        int64_t mode,
+        c10::optional<torch::Tensor> td_in_opt = c10::nullopt) {
     CHECK_IN(dY); CHECK_IN(dHpre); CHECK_IN(levels); CHECK_IN(Hact);
     const int64_t B = levels.size(0), N = levels.size(1),
                   L = levels.size(2), d = levels.size(3);
@@ -831,12 +844,17 @@ std::vector<torch::Tensor> ff_bwd_dw(
     hipStream_t s = cur_stream();
     torch::Tensor td_in;
     if (mode == 1) {
-        auto pos = pos_opt.value();
-        CHECK_IN(pos);
-        td_in = torch::empty({B, N, G, d}, opts);
-        launch_add_pos(levels.data_ptr(), pos.data_ptr(), td_in.data_ptr(),
-                       td_in.numel(), (int)N, (int)L, (int)d, s);
-        check_launch();
+        if (td_in_opt.has_value() && td_in_opt->numel() > 0) {
+            td_in = td_in_opt.value();   // saved by the forward
+        } else {
+            auto pos = pos_opt.value();
+            CHECK_IN(pos);
+            td_in = torch::empty({B, N, G, d}, opts);
+            launch_add_pos(levels.data_ptr(), pos.data_ptr(),
+                           td_in.data_ptr(), td_in.numel(), (int)N, (int)L,
+                           (int)d, s);
+            check_launch();
+        }
     }
     auto dW1 = torch::empty({G * m4, d}, opts);
     {
@@ -976,8 +994,12 @@ torch::Tensor dpos(torch::Tensor dLevels) {
     const int64_t B = dLevels.size(0), N = dLevels.size(1),
                   L = dLevels.size(2), d = dLevels.size(3);
     auto out = torch::empty({N, d}, dLevels.options());
-    launch_dpos(dLevels.data_ptr(), out.data_ptr(), (int)B, (int)N, (int)L,
-                (int)d, cur_stream());
+    const int BCH = (int)std::min<int64_t>(B, 8);
+    auto ws = torch::empty({BCH * N * d},
+                           dLevels.options().dtype(at::kFloat));
+    launch_dpos(dLevels.data_ptr(), out.data_ptr(),
+                d % 8 == 0 ? ws.data_ptr<float>() : nullptr, BCH,
+                (int)B, (int)N, (int)L, (int)d, cur_stream());
     check_launch();
     return out;
 }
@@ -1050,7 +1072,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("pos"), py::arg("w1"), py::arg("w2"),
           py::arg("Hpre"), py::arg("Hact"), py::arg("mode"),
           py::arg("w1t") = c10::nullopt,
-          py::arg("w2t") = c10::nullopt);
+          py::arg("w2t") = c10::nullopt,
+          py::arg("td_in") = c10::nullopt);
     m.def("consensus_fwd", &consensus_fwd, "consensus attention forward");
     m.def("consensus_bwd", &consensus_bwd, "consensus attention backward");
     m.def("level_mix_fwd", &level_mix_fwd, "level mix forward",
@@ -1063,7 +1086,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("tw1"), py::arg("tb1"), py::arg("tw2"), py::arg("tb2"),
           py::arg("attend_self"), py::arg("mask"),
           py::arg("slab") = c10::nullopt, py::arg("slab_idx") = 0);
-    m.def("glom_step_bwd", &glom_step_bwd, "full GLOM iteration backward");
+    m.def("glom_step_bwd", &glom_step_bwd, "full GLOM iteration backward",
+          py::arg("dnew"), py::arg("tokens"), py::arg("levels"),
+          py::arg("pos"), py::arg("bw1"), py::arg("bw2"), py::arg("tw1"),
+          py::arg("tw2"), py::arg("buHpre"), py::arg("buHact"),
+          py::arg("tdHpre"), py::arg("tdHact"), py::arg("probs"),
+          py::arg("rnorm"), py::arg("attend_self"), py::arg("mask"),
+          py::arg("bw1t") = c10::nullopt, py::arg("bw2t") = c10::nullopt,
+          py::arg("tw1t") = c10::nullopt, py::arg("tw2t") = c10::nullopt,
+          py::arg("td_in") = c10::nullopt);
     m.def("patch_embed_fwd", &patch_embed_fwd,
           "patchify + embed GEMM (K1)");
     m.def("patch_embed_bwd", &patch_embed_bwd, "patch embed backward");
@@ -1073,7 +1104,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("add4_into", &add4_into, "fused 4-way elementwise sum");
     m.def("ff_bwd_dh", &ff_bwd_dh);
     m.def("ff_bwd_dx", &ff_bwd_dx);
-    m.def("ff_bwd_dw", &ff_bwd_dw);
+    m.def("ff_bwd_dw", &ff_bwd_dw, py::arg("dY"), py::arg("dHpre"),
+          py::arg("tokens"), py::arg("levels"), py::arg("pos"),
+          py::arg("Hact"), py::arg("mode"),
+          py::arg("td_in") = c10::nullopt);
     m.def("set_nt8p", &set_nt8p, "toggle the 8-phase NT kernel (A/B)");
     m.def("build_info", &build_info);
     m.def("bench_gemm", &bench_gemm, "raw GEMM microbench (tuning only)");

@@ -18,9 +18,11 @@ void launch_slice_cols(const void* src, void* dst, long rows, int Kp,
 int colsum_rb(long M);
 void launch_colsum(const void* in, float* partials, void* out, int nprob,
                    long M, long C, hipStream_t s);
-// dLevels (B,N,L,d) -> dPos (N,d) = sum over b and l in [1,L)
-void launch_dpos(const void* dlev, void* out, int B, int N, int L, int d,
-                 hipStream_t s);
+// dLevels (B,N,L,d) -> dPos (N,d) = sum over b and l in [1,L).
+// partials: caller-provided (BCH, N, d) f32 workspace (BCH batch chunks);
+// pass nullptr/0 for the scalar single-stage fallback.
+void launch_dpos(const void* dlev, void* out, float* partials, int BCH,
+                 int B, int N, int L, int d, hipStream_t s);
 // zero the single level slice out[:, :, l0, :] of a (B,N,L,d) tensor
 void launch_zero_slice(void* out, long BN, int L, int d, int l0,
                        hipStream_t s);

@@ -130,22 +130,37 @@ int colsum_rb(long M) {
     return (int)(rb < 256 ? rb : 256);
 }
 
-// stage 1: each block owns (problem p, row-chunk rc, 256-column tile) and
-// sums its chunk's rows into partials[(p*RB + rc)*C + c]; reads are
-// coalesced along C.
+// stage 1: each block owns (problem p, row-chunk rc, column tile) and
+// sums its chunk's rows into partials[(p*RB + rc)*C + c]. 8-wide 16B
+// loads when C % 8 == 0 (every model shape); scalar fallback otherwise.
 __global__ __launch_bounds__(NT256) void k_colsum_part(
         const ushort_t* __restrict__ in, float* __restrict__ partials,
         long M, long C, int RB) {
-    int c = blockIdx.x * NT256 + threadIdx.x;
     int rc = blockIdx.y;
     int p = blockIdx.z;
-    if (c >= C) return;
     long chunk = cdivl(M, RB);
     long r0 = rc * chunk, r1 = min((long)M, r0 + chunk);
     const ushort_t* base = in + (long)p * M * C;
+    float* pout = partials + ((long)p * RB + rc) * C;
+    if (C % 8 == 0) {
+        long c8 = ((long)blockIdx.x * NT256 + threadIdx.x) * 8;
+        if (c8 >= C) return;
+        float acc[8] = {};
+        for (long r = r0; r < r1; r++) {
+            union { uint4v v; ushort_t u[8]; } t;
+            t.v = *(const uint4v*)(base + r * C + c8);
+#pragma unroll
+            for (int e = 0; e < 8; e++) acc[e] += bf2f(t.u[e]);
+        }
+#pragma unroll
+        for (int e = 0; e < 8; e++) pout[c8 + e] = acc[e];
+        return;
+    }
+    long c = (long)blockIdx.x * NT256 + threadIdx.x;
+    if (c >= C) return;
     float acc = 0.f;
     for (long r = r0; r < r1; r++) acc += bf2f(base[r * C + c]);
-    partials[((long)p * RB + rc) * C + c] = acc;
+    pout[c] = acc;
 }
 
 // stage 2: out[p][c] = bf16(sum_rc partials[p][rc][c])
@@ -164,8 +179,9 @@ __global__ __launch_bounds__(NT256) void k_colsum_fin(
 void launch_colsum(const void* in, float* partials, void* out, int nprob,
                    long M, long C, hipStream_t s) {
     int RB = colsum_rb(M);
+    long ctiles = (C % 8 == 0) ? cdivl(C / 8, NT256) : cdivl(C, NT256);
     hipLaunchKernelGGL(k_colsum_part,
-                       dim3(cdivl(C, NT256), RB, nprob), dim3(NT256), 0, s,
+                       dim3(ctiles, RB, nprob), dim3(NT256), 0, s,
                        (const ushort_t*)in, partials, M, C, RB);
     hipLaunchKernelGGL(k_colsum_fin, dim3(cdivl(C, NT256), nprob),
                        dim3(NT256), 0, s, partials, (ushort_t*)out, C, RB);
@@ -175,6 +191,46 @@ void launch_colsum(const void* in, float* partials, void* out, int nprob,
 // positional-embedding gradient: dPos[n, q] = sum_b sum_{l=1..L-1}
 // dLevels[b, n, l, q]  (deterministic: fixed b-then-l order per thread)
 
+// two-stage when B is large: stage A sums a batch chunk into f32
+// partials (grid.z = BCH chunks -> fills the chip), stage B folds the
+// chunks and writes bf16. d % 8 == 0 on every native config.
+__global__ __launch_bounds__(NT256) void k_dpos_part(
+        const ushort_t* __restrict__ dlev, float* __restrict__ partials,
+        int B, int N, int L, int d, int BCH) {
+    long q8 = ((long)blockIdx.x * NT256 + threadIdx.x) * 8;
+    int n = blockIdx.y;
+    int bc = blockIdx.z;
+    if (q8 >= d) return;
+    int bchunk = (B + BCH - 1) / BCH;
+    int b0 = bc * bchunk, b1 = min(B, b0 + bchunk);
+    float acc[8] = {};
+    const long bstride = (long)N * L * d;
+    const ushort_t* base = dlev + ((long)n * L + 1) * d + q8;
+    for (int b = b0; b < b1; b++) {
+        const ushort_t* p = base + b * bstride;
+        for (int l = 0; l < L - 1; l++) {
+            union { uint4v v; ushort_t u[8]; } t;
+            t.v = *(const uint4v*)(p + (long)l * d);
+#pragma unroll
+            for (int e = 0; e < 8; e++) acc[e] += bf2f(t.u[e]);
+        }
+    }
+    float* pout = partials + ((long)bc * N + n) * d + q8;
+#pragma unroll
+    for (int e = 0; e < 8; e++) pout[e] = acc[e];
+}
+
+__global__ __launch_bounds__(NT256) void k_dpos_fin(
+        const float* __restrict__ partials, ushort_t* __restrict__ out,
+        long Nd, int BCH) {
+    long i = (long)blockIdx.x * NT256 + threadIdx.x;
+    if (i >= Nd) return;
+    float acc = 0.f;
+    for (int bc = 0; bc < BCH; bc++) acc += partials[(long)bc * Nd + i];
+    out[i] = f2bf(acc);
+}
+
+// scalar fallback for d % 8 != 0 (not reachable from the native path)
 __global__ __launch_bounds__(NT256) void k_dpos(
         const ushort_t* __restrict__ dlev, ushort_t* __restrict__ out,
         int B, int N, int L, int d) {
@@ -191,8 +247,18 @@ __global__ __launch_bounds__(NT256) void k_dpos(
     out[(long)n * d + q] = f2bf(acc);
 }
 
-void launch_dpos(const void* dlev, void* out, int B, int N, int L, int d,
-                 hipStream_t s) {
+void launch_dpos(const void* dlev, void* out, float* partials, int BCH,
+                 int B, int N, int L, int d, hipStream_t s) {
+    if (d % 8 == 0 && partials != nullptr && BCH > 0) {
+        hipLaunchKernelGGL(k_dpos_part,
+                           dim3(cdivl(d / 8, NT256), N, BCH), dim3(NT256),
+                           0, s, (const ushort_t*)dlev, partials, B, N, L,
+                           d, BCH);
+        hipLaunchKernelGGL(k_dpos_fin,
+                           dim3(cdivl((long)N * d, NT256)), dim3(NT256), 0,
+                           s, partials, (ushort_t*)out, (long)N * d, BCH);
+        return;
+    }
     hipLaunchKernelGGL(k_dpos, dim3(cdivl(d, NT256), N), dim3(NT256), 0, s,
                        (const ushort_t*)dlev, (ushort_t*)out, B, N, L, d);
 }

@@ -72,19 +72,20 @@ class GroupedFFFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, tokens, levels, pos, w1, b1, w2, b2, mode):
         ext = _load_extension()
-        Y, Hpre, Hact = ext.grouped_ff_fwd(tokens, levels, pos, w1, b1, w2,
-                                           b2, mode)
-        ctx.save_for_backward(tokens, levels, pos, w1, w2, Hpre, Hact)
+        Y, Hpre, Hact, td_in = ext.grouped_ff_fwd(tokens, levels, pos, w1,
+                                                  b1, w2, b2, mode)
+        ctx.save_for_backward(tokens, levels, pos, w1, w2, Hpre, Hact,
+                              td_in)
         ctx.mode = mode
         return Y
 
     @staticmethod
     def backward(ctx, dY):
         ext = _load_extension()
-        tokens, levels, pos, w1, w2, Hpre, Hact = ctx.saved_tensors
+        tokens, levels, pos, w1, w2, Hpre, Hact, td_in = ctx.saved_tensors
         dTokens, dLevels, dW1, dB1, dW2, dB2 = ext.grouped_ff_bwd(
             dY.contiguous(), tokens, levels, pos, w1, w2, Hpre, Hact,
-            ctx.mode)
+            ctx.mode, None, None, td_in)
         dPos = None
         if ctx.mode == 1 and pos is not None and ctx.needs_input_grad[2]:
             # pos was added to every top-down group input; its grad is the
@@ -160,7 +161,8 @@ class GlomStepFn(torch.autograd.Function):
         slab, sidx = slab_ref if slab_ref is not None else (None, 0)
         if (torch.is_grad_enabled()
                 and os.environ.get("GLOM_FWD_STREAMS", "0") != "1"):
-            out, bhp, bha, thp, tha, probs, rnorm = ext.glom_step_fwd(
+            (out, bhp, bha, thp, tha, probs, rnorm,
+             tdin) = ext.glom_step_fwd(
                 tokens, levels, pos, bw1, bb1, bw2, bb2, tw1, tb1, tw2,
                 tb2, attend_self, mask, slab, sidx)
         else:
@@ -173,12 +175,13 @@ class GlomStepFn(torch.autograd.Function):
             s_td, s_at, _ = GlomStepFn._side_streams()
             ev = torch.cuda.Event()
             ev.record(cur)
-            buY, bhp, bha = ext.grouped_ff_fwd(tokens, levels, None,
-                                               bw1, bb1, bw2, bb2, 0)
+            buY, bhp, bha, _ = ext.grouped_ff_fwd(tokens, levels, None,
+                                                  bw1, bb1, bw2, bb2, 0)
             with torch.cuda.stream(s_td):
                 s_td.wait_event(ev)
-                tdY, thp, tha = ext.grouped_ff_fwd(None, levels, pos,
-                                                   tw1, tb1, tw2, tb2, 1)
+                tdY, thp, tha, tdin = ext.grouped_ff_fwd(None, levels, pos,
+                                                         tw1, tb1, tw2,
+                                                         tb2, 1)
             with torch.cuda.stream(s_at):
                 s_at.wait_event(ev)
                 cons, probs, rnorm = ext.consensus_fwd(levels, attend_self,
@@ -187,12 +190,12 @@ class GlomStepFn(torch.autograd.Function):
             cur.wait_stream(s_at)
             # boundary tensors crossing back to the main stream: pin their
             # blocks until the main stream catches up (allocator safety)
-            for t in (tdY, thp, tha, cons, probs, rnorm):
+            for t in (tdY, thp, tha, tdin, cons, probs, rnorm):
                 t.record_stream(cur)
             out = ext.level_mix_fwd(levels, buY, tdY, cons, slab, sidx)
         ctx.save_for_backward(tokens, levels, pos, bw1, bw2, tw1, tw2,
                               bhp, bha, thp, tha, probs, rnorm, mask,
-                              bw1t, bw2t, tw1t, tw2t)
+                              bw1t, bw2t, tw1t, tw2t, tdin)
         ctx.attend_self = attend_self
         return out
 
@@ -200,13 +203,14 @@ class GlomStepFn(torch.autograd.Function):
     def backward(ctx, dnew):
         ext = _load_extension()
         (tokens, levels, pos, bw1, bw2, tw1, tw2, bhp, bha, thp, tha,
-         probs, rnorm, mask, bw1t, bw2t, tw1t, tw2t) = ctx.saved_tensors
+         probs, rnorm, mask, bw1t, bw2t, tw1t, tw2t,
+         tdin) = ctx.saved_tensors
         if os.environ.get("GLOM_NO_BWD_STREAMS", "0") == "1":
             (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
              dtw1, dtb1, dtw2, dtb2) = ext.glom_step_bwd(
                 dnew.contiguous(), tokens, levels, pos, bw1, bw2, tw1, tw2,
                 bhp, bha, thp, tha, probs, rnorm, ctx.attend_self, mask,
-                bw1t, bw2t, tw1t, tw2t)
+                bw1t, bw2t, tw1t, tw2t, tdin)
             return (dTokens, dLevels, dPos, dbw1, dbb1, dbw2, dbb2,
                     dtw1, dtb1, dtw2, dtb2, None, None, None, None, None,
                     None, None)
@@ -221,7 +225,7 @@ class GlomStepFn(torch.autograd.Function):
             with torch.cuda.stream(s_td):
                 s_td.wait_event(ev)
                 td = ext.grouped_ff_bwd(dtd, None, levels, pos, tw1, tw2,
-                                        thp, tha, 1, tw1t, tw2t)
+                                        thp, tha, 1, tw1t, tw2t, tdin)
             with torch.cuda.stream(s_at):
                 s_at.wait_event(ev)
                 dAttn = ext.consensus_bwd(dmix, levels, probs, rnorm,
@@ -266,7 +270,7 @@ class GlomStepFn(torch.autograd.Function):
                                                 levels, None, bha, 0)
             s_w.wait_event(ev_td)
             td_w1, td_w2, td_b2 = ext.ff_bwd_dw(dtd, td_dh, None, levels,
-                                                pos, tha, 1)
+                                                pos, tha, 1, tdin)
         bu_dt, bu_dl = ext.ff_bwd_dx(bu_dh, bw1t, tokens, B, N, L, 0)
         with torch.cuda.stream(s_td):
             _, td_dl = ext.ff_bwd_dx(td_dh, tw1t, None, B, N, L, 1)
