@@ -193,7 +193,14 @@ class Glom(nn.Module):
 
     def forward(self, img: torch.Tensor, iters: int | None = None,
                 levels: torch.Tensor | None = None,
-                return_all: bool = False) -> torch.Tensor:
+                return_all: bool = False,
+                grad_iters: int | None = None) -> torch.Tensor:
+        """grad_iters (extension kwarg, default None = exact reference
+        autograd graph): run iterations >= grad_iters under no_grad.
+        Forward values are unchanged; use when the loss only reads
+        trajectory times <= grad_iters (e.g. the denoising recipe) to skip
+        backprop through post-loss iterations whose gradient contribution
+        is exactly zero."""
         if img.dim() != 4 or img.shape[1] != 3:
             raise ValueError(f"expected image batch (b, 3, H, W), got {tuple(img.shape)}")
         iters = iters if iters is not None else 2 * self.levels
@@ -206,8 +213,10 @@ class Glom(nn.Module):
                     return cache.run(img, iters, levels, return_all)
             from glom_pytorch_amd.ops import native_forward
             return native_forward(self, img, iters=iters, levels=levels,
-                                  return_all=return_all)
-        return self._eager_forward(img, iters, levels, return_all)
+                                  return_all=return_all,
+                                  grad_iters=grad_iters)
+        return self._eager_forward(img, iters, levels, return_all,
+                                   grad_iters)
 
     def enable_graphs(self):
         """hipGraph-capture the T-step loop for inference: each distinct
@@ -227,7 +236,8 @@ class Glom(nn.Module):
     # compare this against an independent loop-level oracle, and GPU tests
     # compare the HIP engine against it.
 
-    def _eager_forward(self, img, iters, levels, return_all):
+    def _eager_forward(self, img, iters, levels, return_all,
+                       grad_iters=None):
         b = img.shape[0]
         tokens = self.image_to_tokens(img)
         n = tokens.shape[1]
@@ -240,8 +250,12 @@ class Glom(nn.Module):
             levels = repeat(self.init_levels, "l d -> b n l d", b=b, n=n)
 
         trajectory = [levels]
-        for _ in range(iters):
-            levels = self._eager_step(bottom, levels, pos)
+        for t in range(iters):
+            if grad_iters is not None and t >= grad_iters:
+                with torch.no_grad():
+                    levels = self._eager_step(bottom, levels, pos)
+            else:
+                levels = self._eager_step(bottom, levels, pos)
             trajectory.append(levels)
 
         if return_all:

@@ -48,8 +48,9 @@ def available() -> bool:
         return False
 
 
-def native_forward(model, img, iters, levels=None, return_all=False):
+def native_forward(model, img, iters, levels=None, return_all=False,
+                   grad_iters=None):
     """Run Glom.forward on the CDNA4 HIP engine (bf16, gfx950)."""
     from glom_pytorch_amd.ops.functional import glom_forward
     return glom_forward(model, img, iters=iters, levels=levels,
-                        return_all=return_all)
+                        return_all=return_all, grad_iters=grad_iters)

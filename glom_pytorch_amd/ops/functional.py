@@ -322,7 +322,15 @@ def glom_step(model, tokens, levels, pos, mask, wts=None, slab_ref=None):
         model.attention.attend_self, mask, *wts, slab_ref)
 
 
-def glom_forward(model, img, iters, levels=None, return_all=False):
+def glom_forward(model, img, iters, levels=None, return_all=False,
+                 grad_iters=None):
+    """grad_iters (optional): iterations >= grad_iters run under no_grad.
+    Forward VALUES are identical; gradients stop flowing through those
+    steps. When a loss only touches trajectory times <= grad_iters (the
+    reference's denoising recipe decodes at t=7 of 12), the skipped
+    backward contributions are exactly zero — autograd would otherwise
+    backprop zeros through every post-loss iteration (reference
+    glom_pytorch.py:131-148 has the same dead work)."""
     from glom_pytorch_amd.utils.profiling import trace_range
     b = img.shape[0]
     with trace_range("glom/patch_embed"):
@@ -355,10 +363,15 @@ def glom_forward(model, img, iters, levels=None, return_all=False):
     steps = [levels]
     with trace_range(f"glom/iterate x{iters}"):
         for t in range(iters):
+            sref = (slab, t + 1) if slab is not None else None
             with trace_range(f"glom/step{t}"):
-                levels = glom_step(
-                    model, tokens, levels, pos, mask, wts,
-                    (slab, t + 1) if slab is not None else None)
+                if grad_iters is not None and t >= grad_iters:
+                    with torch.no_grad():
+                        levels = glom_step(model, tokens, levels, pos,
+                                           mask, wts, sref)
+                else:
+                    levels = glom_step(model, tokens, levels, pos, mask,
+                                       wts, sref)
             if return_all:
                 steps.append(levels)
 

@@ -128,7 +128,11 @@ class DenoisingTrainer:
         t = min(self.decode_step, iters)
         self.opt.zero_grad(set_to_none=True)
         noised = img + torch.randn_like(img) * self.noise_std
-        all_levels = self.model(noised, iters=iters, return_all=True)
+        # the loss reads only trajectory time t: iterations >= t carry
+        # exactly-zero gradient, so they run forward-only (grad_iters) —
+        # identical values and gradients, ~(iters-t)/iters less backward
+        all_levels = self.model(noised, iters=iters, return_all=True,
+                                grad_iters=t)
         top = all_levels[t, :, :, -1]
         recon = self.decoder(top)
         loss = F.mse_loss(recon.float(), img.float())

@@ -148,3 +148,37 @@ def test_native_capability_envelope():
     for mod in (m, m2, m3):
         out = mod(img, iters=1)
         assert out.shape == (1, 16, mod.levels, mod.dim)
+
+
+def test_grad_iters_gradients_identical():
+    """grad_iters runs post-loss iterations forward-only; when the loss
+    reads only trajectory times <= grad_iters the gradients are EXACTLY
+    the ones of the full autograd graph (the skipped contributions are
+    structurally zero)."""
+    import torch
+    from glom_pytorch_amd import Glom
+
+    torch.manual_seed(0)
+    m = Glom(dim=32, levels=3, image_size=16, patch_size=8)
+    img = torch.randn(2, 3, 16, 16)
+
+    def run(grad_iters):
+        for p in m.parameters():
+            p.grad = None
+        traj = m(img, iters=4, return_all=True, grad_iters=grad_iters)
+        loss = traj[2].pow(2).mean()
+        loss.backward()
+        return (loss.detach().clone(),
+                {n: p.grad.clone() for n, p in m.named_parameters()
+                 if p.grad is not None})
+
+    l_full, g_full = run(None)
+    l_cut, g_cut = run(2)
+    assert torch.equal(l_full, l_cut)
+    assert set(g_full) == set(g_cut)
+    for n in g_full:
+        assert torch.equal(g_full[n], g_cut[n]), n
+    # forward values of the skipped iterations are still produced
+    traj = m(img, iters=4, return_all=True, grad_iters=2)
+    assert traj.shape[0] == 5
+    assert torch.isfinite(traj).all()
