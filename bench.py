@@ -147,6 +147,12 @@ def main():
                 "objective": ("denoising-mse fwd+bwd+adamw"
                               if args.mode == "train"
                               else "inference forward (hipGraph replay)"),
+                # reference recipe (README.md:56-90): decode the top level
+                # at t=7; iterations >= 7 run forward-only since their
+                # gradient contribution is exactly zero (dead-graph
+                # elimination, gradients bitwise-identical; applied to the
+                # eager/compile comparison arms too)
+                "decode_step": 7,
                 "impl": args.impl,
                 "parallelism": f"dp{n_gpus}",
             },
