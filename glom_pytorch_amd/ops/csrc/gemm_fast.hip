@@ -1234,6 +1234,10 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast5p_kernel(GemmParams p) {
     float* colp = nullptr;
     if (p.colsum_out)
         colp = p.colsum_out + (long)(pid % p.nInner) * p.colsum_sin;
+    ushort_t* o2p = nullptr;   // EPI_GELU_PAIR: gelu(pre-act) output
+    if (p.epilogue == EPI_GELU_PAIR && p.out2)
+        o2p = (ushort_t*)p.out2 + (long)(pid % p.nInner) * p.out2_sin
+              + (long)(pid / p.nInner) * p.out2_sout;
 
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
@@ -1360,6 +1364,7 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast5p_kernel(GemmParams p) {
                         for (int j16 = 0; j16 < 4; j16++) vv[j16] *= gy[j16];
                     }
                     ushort_t* crow = Cp + gi * ldc + n0 + wn + lrow;
+                    ushort_t cbv[4];
 #pragma unroll
                     for (int j16 = 0; j16 < 4; j16++) {
                         ushort_t cb = f2bf(vv[j16] + bvv[j16]);
@@ -1368,7 +1373,22 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast5p_kernel(GemmParams p) {
                         // per summand (only the add order differs, and
                         // that was already atomic-nondeterministic)
                         colacc[j16] += bf2f(cb);
+                        cbv[j16] = cb;
                         crow[j16 * 16] = cb;
+                    }
+                    if (o2p) {
+                        // gelu of the bf16-rounded pre-activation:
+                        // bitwise the standalone k_gelu pass
+                        float gx[4], gy[4];
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            gx[j16] = bf2f(cbv[j16]);
+                        gelu_f_vec<4>(gx, gy);
+                        ushort_t* orow = o2p + gi * p.out2_ld + n0 + wn
+                                         + lrow;
+#pragma unroll
+                        for (int j16 = 0; j16 < 4; j16++)
+                            orow[j16 * 16] = f2bf(gy[j16]);
                     }
                     acc[i16][0][r] = 0.f; acc[i16][1][r] = 0.f;
                     acc[i16][2][r] = 0.f; acc[i16][3][r] = 0.f;
