@@ -213,7 +213,11 @@ class DenoisingTrainer:
             for q in self._params:
                 q.grad = None
             graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
+            # thread_local error mode: the RCCL watchdog thread polls
+            # events on this device during capture; global mode would
+            # invalidate the capture for those unrelated API calls
+            with torch.cuda.graph(graph,
+                                  capture_error_mode="thread_local"):
                 loss = self._eager_step(static_img, iters)
             # capture RECORDS the step without executing it: undo the
             # Python-side count bump from the captured call
