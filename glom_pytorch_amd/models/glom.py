@@ -194,13 +194,18 @@ class Glom(nn.Module):
     def forward(self, img: torch.Tensor, iters: int | None = None,
                 levels: torch.Tensor | None = None,
                 return_all: bool = False,
-                grad_iters: int | None = None) -> torch.Tensor:
+                grad_iters: int | None = None,
+                overlap_tail: bool = False) -> torch.Tensor:
         """grad_iters (extension kwarg, default None = exact reference
         autograd graph): run iterations >= grad_iters under no_grad.
         Forward values are unchanged; use when the loss only reads
         trajectory times <= grad_iters (e.g. the denoising recipe) to skip
         backprop through post-loss iterations whose gradient contribution
-        is exactly zero."""
+        is exactly zero. ``overlap_tail`` (native path only) additionally
+        runs those iterations on a side stream so they overlap the
+        caller's backward; the caller MUST then call
+        ``ops.functional.join_tail_stream()`` before reading the tail's
+        trajectory slices or mutating weights."""
         if img.dim() != 4 or img.shape[1] != 3:
             raise ValueError(f"expected image batch (b, 3, H, W), got {tuple(img.shape)}")
         iters = iters if iters is not None else 2 * self.levels
@@ -214,7 +219,8 @@ class Glom(nn.Module):
             from glom_pytorch_amd.ops import native_forward
             return native_forward(self, img, iters=iters, levels=levels,
                                   return_all=return_all,
-                                  grad_iters=grad_iters)
+                                  grad_iters=grad_iters,
+                                  overlap_tail=overlap_tail)
         return self._eager_forward(img, iters, levels, return_all,
                                    grad_iters)
 

@@ -49,8 +49,9 @@ def available() -> bool:
 
 
 def native_forward(model, img, iters, levels=None, return_all=False,
-                   grad_iters=None):
+                   grad_iters=None, overlap_tail=False):
     """Run Glom.forward on the CDNA4 HIP engine (bf16, gfx950)."""
     from glom_pytorch_amd.ops.functional import glom_forward
     return glom_forward(model, img, iters=iters, levels=levels,
-                        return_all=return_all, grad_iters=grad_iters)
+                        return_all=return_all, grad_iters=grad_iters,
+                        overlap_tail=overlap_tail)

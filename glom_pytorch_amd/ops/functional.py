@@ -322,8 +322,27 @@ def glom_step(model, tokens, levels, pos, mask, wts=None, slab_ref=None):
         model.attention.attend_self, mask, *wts, slab_ref)
 
 
+_TAIL_STREAM = None
+
+
+def _tail_stream():
+    global _TAIL_STREAM
+    if _TAIL_STREAM is None:
+        _TAIL_STREAM = torch.cuda.Stream()
+    return _TAIL_STREAM
+
+
+def join_tail_stream():
+    """Block the current stream on the overlap-tail stream. Callers that
+    requested ``overlap_tail`` MUST call this before reading the
+    trajectory's post-grad_iters slices or mutating model weights (the
+    trainer joins right before the optimizer update)."""
+    if _TAIL_STREAM is not None:
+        torch.cuda.current_stream().wait_stream(_TAIL_STREAM)
+
+
 def glom_forward(model, img, iters, levels=None, return_all=False,
-                 grad_iters=None):
+                 grad_iters=None, overlap_tail=False):
     """grad_iters (optional): iterations >= grad_iters run under no_grad.
     Forward VALUES are identical; gradients stop flowing through those
     steps. When a loss only touches trajectory times <= grad_iters (the
@@ -360,9 +379,18 @@ def glom_forward(model, img, iters, levels=None, return_all=False,
         with torch.no_grad():
             slab[0].copy_(levels)
 
+    # opt-in: run the forward-only tail (iterations >= grad_iters)
+    # CONCURRENTLY with the caller's backward on a dedicated stream. The
+    # tail never feeds the loss, and the caller joins (join_tail_stream)
+    # before the optimizer mutates weights, so this is race-free.
+    overlap = (overlap_tail and grad_iters is not None
+               and grad_iters < iters and img.is_cuda
+               and torch.is_grad_enabled())
+    n_sync = grad_iters if overlap else iters
+
     steps = [levels]
     with trace_range(f"glom/iterate x{iters}"):
-        for t in range(iters):
+        for t in range(n_sync):
             sref = (slab, t + 1) if slab is not None else None
             with trace_range(f"glom/step{t}"):
                 if grad_iters is not None and t >= grad_iters:
@@ -374,6 +402,28 @@ def glom_forward(model, img, iters, levels=None, return_all=False,
                                        wts, sref)
             if return_all:
                 steps.append(levels)
+        if overlap:
+            ext = _load_extension()
+            bw = model.bottom_up.net
+            tw = model.top_down.net
+            s_tail = _tail_stream()
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            with torch.no_grad(), torch.cuda.stream(s_tail):
+                s_tail.wait_event(ev)
+                for t in range(n_sync, iters):
+                    out8 = ext.glom_step_fwd(
+                        tokens, levels, pos,
+                        bw[1].weight[..., 0], bw[1].bias,
+                        bw[3].weight[..., 0], bw[3].bias,
+                        tw[1].weight[..., 0], tw[1].bias,
+                        tw[3].weight[..., 0], tw[3].bias,
+                        model.attention.attend_self, mask, slab, t + 1)
+                    levels = out8[0]
+                    if return_all:
+                        steps.append(levels)
+                if slab is not None:
+                    slab.record_stream(s_tail)
 
     if return_all:
         return TrajectoryFn.apply(slab, *steps)

@@ -130,9 +130,14 @@ class DenoisingTrainer:
         noised = img + torch.randn_like(img) * self.noise_std
         # the loss reads only trajectory time t: iterations >= t carry
         # exactly-zero gradient, so they run forward-only (grad_iters) —
-        # identical values and gradients, ~(iters-t)/iters less backward
+        # identical values and gradients, ~(iters-t)/iters less backward.
+        # overlap_tail additionally runs them on a side stream so they
+        # overlap the backward below; join_tail_stream() before the
+        # optimizer guarantees no weight write races the tail's reads.
+        overlap = img.is_cuda and not getattr(self.model, "force_eager",
+                                              False)
         all_levels = self.model(noised, iters=iters, return_all=True,
-                                grad_iters=t)
+                                grad_iters=t, overlap_tail=overlap)
         top = all_levels[t, :, :, -1]
         recon = self.decoder(top)
         loss = F.mse_loss(recon.float(), img.float())
@@ -140,6 +145,9 @@ class DenoisingTrainer:
         if self.distributed:
             self.ddp_model.finalize()
             self.ddp_dec.finalize()
+        if overlap:
+            from glom_pytorch_amd.ops.functional import join_tail_stream
+            join_tail_stream()
         if self.fused_opt is not None:
             self.fused_opt.step()
         elif self.master is not None:
