@@ -35,10 +35,22 @@ class _GraphEntry:
             torch.cuda.current_stream().wait_stream(s)
 
             self.graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self.graph):
-                self.static_out = glom_forward(model, self.static_img, iters,
-                                               levels=self.static_levels,
-                                               return_all=return_all)
+            # flush pending destructors and disable cyclic GC during
+            # capture: a stale graph/tensor destructor firing mid-capture
+            # invalidates it (replay would segfault)
+            import gc
+            gc.collect()
+            was = gc.isenabled()
+            gc.disable()
+            try:
+                with torch.cuda.graph(self.graph,
+                                      capture_error_mode="thread_local"):
+                    self.static_out = glom_forward(
+                        model, self.static_img, iters,
+                        levels=self.static_levels, return_all=return_all)
+            finally:
+                if was:
+                    gc.enable()
 
     def replay(self, img, levels):
         self.static_img.copy_(img)

@@ -221,12 +221,23 @@ class DenoisingTrainer:
             for q in self._params:
                 q.grad = None
             graph = torch.cuda.CUDAGraph()
-            # thread_local error mode: the RCCL watchdog thread polls
-            # events on this device during capture; global mode would
-            # invalidate the capture for those unrelated API calls
-            with torch.cuda.graph(graph,
-                                  capture_error_mode="thread_local"):
-                loss = self._eager_step(static_img, iters)
+            # (a) flush pending destructors NOW and keep the cyclic GC off
+            # during capture: a stale CUDAGraph/tensor destructor firing
+            # mid-capture makes HIP API calls that invalidate the capture
+            # (the replay then segfaults); (b) thread_local error mode:
+            # the RCCL watchdog thread polls events on this device during
+            # capture and must not invalidate it either.
+            import gc
+            gc.collect()
+            gc_was_enabled = gc.isenabled()
+            gc.disable()
+            try:
+                with torch.cuda.graph(graph,
+                                      capture_error_mode="thread_local"):
+                    loss = self._eager_step(static_img, iters)
+            finally:
+                if gc_was_enabled:
+                    gc.enable()
             # capture RECORDS the step without executing it: undo the
             # Python-side count bump from the captured call
             fo._step_host = max(0, fo._step_host - 1)
