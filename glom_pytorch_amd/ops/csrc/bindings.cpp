@@ -42,15 +42,15 @@ hipStream_t cur_stream() {
     return c10::hip::getCurrentHIPStream().stream();
 }
 
-// nt8p dispatch mode: opt-in (GLOM_NT8P=1), runtime-settable for
-// within-process A/B microbenches (guide §5.4 rule 24). Measured: +15%
-// on the down-projection microbench but -1% END-TO-END — the 128 KiB
-// LDS / 1-block-per-CU tile starves the concurrently running backward
-// streams of co-resident blocks, so the step's 2-stream overlap loses
-// more than the kernel gains (profiles/README.md round-2 notes).
+// nt8p dispatch mode: default ON for K>=2048 shapes (GLOM_NT8P=0 opts
+// out; set_nt8p() for within-process A/B, guide §5.4 rule 24). History:
+// +15% on the down-projection microbench but -1% end-to-end under the
+// pre-overlap stream mix; with the forward-tail/backward overlap the
+// balance flipped and nt8p measures +1.8% e2e (1437 -> 1463 img/s,
+// same box) — see profiles/README.md round-2 notes.
 int g_nt8p_mode = []() {
     const char* e = getenv("GLOM_NT8P");
-    return (e && e[0] == '1') ? 1 : 0;
+    return (e && e[0] == '0') ? 0 : 1;
 }();
 
 void set_nt8p(bool on) { g_nt8p_mode = on ? 1 : 0; }
