@@ -38,7 +38,8 @@ class DenoisingDecoder(nn.Sequential):
 class DenoisingTrainer:
     def __init__(self, model, *, lr=3e-4, noise_std=1.0, decode_step=7,
                  grad_clip=1.0, distributed=False, bucket_bytes=16 << 20,
-                 process_group=None, graph_step=True, log_path=None):
+                 process_group=None, graph_step=True, overlap_tail=True,
+                 log_path=None):
         """``process_group``: the group whose ranks are data-parallel
         replicas (default: the global group). For 2-D DP x SP meshes pass
         the DP subgroup here — BucketedDDP AVERAGES over its group, which
@@ -94,6 +95,9 @@ class DenoisingTrainer:
         self.graph_step = (graph_step
                            and os.environ.get("GLOM_NO_GRAPH_STEP", "0")
                            != "1")
+        self.overlap_tail = (overlap_tail
+                             and os.environ.get("GLOM_NO_OVERLAP_TAIL",
+                                                "0") != "1")
         self._graphs: dict = {}
 
     @staticmethod
@@ -134,8 +138,8 @@ class DenoisingTrainer:
         # overlap_tail additionally runs them on a side stream so they
         # overlap the backward below; join_tail_stream() before the
         # optimizer guarantees no weight write races the tail's reads.
-        overlap = img.is_cuda and not getattr(self.model, "force_eager",
-                                              False)
+        overlap = (self.overlap_tail and img.is_cuda
+                   and not getattr(self.model, "force_eager", False))
         all_levels = self.model(noised, iters=iters, return_all=True,
                                 grad_iters=t, overlap_tail=overlap)
         top = all_levels[t, :, :, -1]

@@ -267,3 +267,24 @@ def test_consensus_parity_large_n(side, dim):
         ref.backward(dO)
         out.backward(dO.to(torch.bfloat16))
         assert _rel_err(lg.grad, l32.grad) < 3e-2, (side, attend_self)
+
+
+def test_overlap_tail_matches_sequential():
+    """Running the forward-only tail concurrently with the backward must
+    not change training: compare two trainers (overlap on/off), same
+    init, noise_std=0, several steps."""
+    img = torch.randn(4, 3, 32, 32, device=DEV).to(torch.bfloat16)
+    ta = _trainer(graph=False, overlap_tail=False)
+    tb = _trainer(graph=False, overlap_tail=True)
+    tb.model.load_state_dict(ta.model.state_dict())
+    tb.decoder.load_state_dict(ta.decoder.state_dict())
+    with torch.no_grad():
+        for mw, q in zip(tb.master, tb._params):
+            mw.copy_(q.float())
+    la = [ta.step(img, iters=4) for _ in range(3)]
+    lb = [tb.step(img, iters=4) for _ in range(3)]
+    for a, b in zip(la, lb):
+        assert abs(a - b) < 1e-3 * max(1.0, abs(a)), (a, b)
+    for (na, pa), (nb, pb) in zip(ta.model.named_parameters(),
+                                  tb.model.named_parameters()):
+        assert torch.equal(pa, pb), na
