@@ -117,8 +117,13 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
     // N >= 1024 only: at N=512 (2 column panels) the persistent sweep
     // measured SLOWER than the 128^2 kernel on the down-projection
     // (637 -> 573 TF), so that shape keeps nt_fast
+    static const long nt5p_ming = []() {
+        const char* e = getenv("GLOM_NT5P_MING");
+        return e ? atol(e) : 256;
+    }();
     const bool nt5p = nt5p_on && nt3 && p.M % 512 == 0
-                      && (long)(p.N / 256) * (p.M / 512) * p.nproblems >= 256
+                      && (long)(p.N / 256) * (p.M / 512) * p.nproblems
+                         >= nt5p_ming
                       && !(p.Cflags & OP_TABLE)
                       && (p.epilogue == EPI_NONE
                           || p.epilogue == EPI_GELUGRAD
@@ -135,9 +140,13 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
     // K >= 2048 (down-projection 685 -> 789 TF, square parity+), while
     // the K=512 up-projection family stays on nt5p's persistent ring
     // (8-phase loses its prologue amortization at 8 K-steps/tile)
+    static const long nt8p_mink = []() {
+        const char* e = getenv("GLOM_NT8P_MINK");
+        return e ? atol(e) : 2048;
+    }();
     const bool nt8p = nt8p_on && p.layout == LAYOUT_NT && lds_ok && no_xform
                       && p.M % 256 == 0 && p.N % 256 == 0 && p.K % 64 == 0
-                      && p.K >= 2048
+                      && p.K >= nt8p_mink
                       && !(p.Cflags & OP_TABLE)
                       && (p.epilogue == EPI_NONE
                           || p.epilogue == EPI_GELUGRAD)
