@@ -117,9 +117,12 @@ void run_gemm(GemmParams& p, hipStream_t s, const torch::TensorOptions& opts,
     // N >= 1024 only: at N=512 (2 column panels) the persistent sweep
     // measured SLOWER than the 128^2 kernel on the down-projection
     // (637 -> 573 TF), so that shape keeps nt_fast
+    // >= 96 (was 256): at the small-batch inference shapes (video B=8:
+    // 192-block grids) the persistent sweep beats nt3 by ~20% (3172 ->
+    // 3812 frames/s); training grids are >= 768 and unaffected.
     static const long nt5p_ming = []() {
         const char* e = getenv("GLOM_NT5P_MING");
-        return e ? atol(e) : 256;
+        return e ? atol(e) : 96;
     }();
     const bool nt5p = nt5p_on && nt3 && p.M % 512 == 0
                       && (long)(p.N / 256) * (p.M / 512) * p.nproblems
