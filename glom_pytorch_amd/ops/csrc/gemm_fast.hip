@@ -987,85 +987,105 @@ __global__ __launch_bounds__(NT3) void gemm_nt_fast4_kernel(GemmParams p) {
                     + (long)(pid % p.nInner) * p.out2_sin
                     + (long)(pid / p.nInner) * p.out2_sout;
                 const int dtot = p.npatch;
-                ushort_t* Vt = smem + 128 * EPI2_ROW;   // [256][FBK]
+                // double-buffered transpose-staged V: stage s = (dc, jt)
+                // loads the NEXT piece before this stage's MFMAs (VMEM in
+                // flight under compute) and transpose-writes it into the
+                // other buffer after them — one barrier per stage instead
+                // of the single-buffered {stage, sync, mfma, sync} chain.
+                ushort_t* Vt0 = smem + 128 * EPI2_ROW;  // [256][FBK] x2
+                ushort_t* Vt1 = Vt0 + 256 * FBK;
                 const int wrow = (wid >> 2) * 64;
                 const int wcol64 = (wid & 3) * 64;
-                for (int dc = 0; dc < dtot; dc += 256) {
-                    f32x4 acc2[4][4] = {};
-                    for (int jt = 0; jt < 4; jt++) {
-                        // transpose-stage V rows jt*64.. cols dc..dc+255
-                        if (threadIdx.x < 256) {
-                            int tt = threadIdx.x;
-                            int cb = tt & 31;
-                            int mb = tt >> 5;
-                            union { uint4v v; ushort_t u[8]; } rv[8];
+                const int nstage = ((dtot + 255) / 256) * 4;
+                const int tt = threadIdx.x, cb = tt & 31, mb = tt >> 5;
+                union Pc { uint4v v; ushort_t u[8]; };
+                Pc rv[8];
+                auto load_piece = [&](int st) {
+                    if (tt < 256) {
+                        int jt = st & 3, dc = (st >> 2) * 256;
 #pragma unroll
-                            for (int r = 0; r < 8; r++) {
-                                int j = jt * 64 + mb * 8 + r;
-                                int gc = dc + cb * 8;
-                                if (gc + 7 < dtot)
-                                    rv[r].v = *(const uint4v*)(Vp
-                                        + (long)j * ldv + gc);
-                                else
-                                    rv[r].v = 0;
-                            }
-#pragma unroll
-                            for (int e = 0; e < 8; e++) {
-                                union { uint4v v; ushort_t u[8]; } col;
-#pragma unroll
-                                for (int r = 0; r < 8; r++)
-                                    col.u[r] = rv[r].u[e];
-                                int row = cb * 8 + e;
-                                int off = (mb * 8) ^ (swz_row(row) << 3);
-                                *(uint4v*)&Vt[row * FBK + off] = col.v;
-                            }
+                        for (int r = 0; r < 8; r++) {
+                            int j = jt * 64 + mb * 8 + r;
+                            int gc = dc + cb * 8;
+                            if (gc + 7 < dtot)
+                                rv[r].v = *(const uint4v*)(Vp
+                                    + (long)j * ldv + gc);
+                            else
+                                rv[r].v = 0;
                         }
-                        __syncthreads();
-                        short8 paf[2][4], vbf[2][4];
-#pragma unroll
-                        for (int ss = 0; ss < 2; ss++) {
-#pragma unroll
-                            for (int i = 0; i < 4; i++) {
-                                int row = wrow + i * 16 + lrow;
-                                int koff = jt * 64 + ss * 32 + kq * 8;
-                                paf[ss][i] = *(const short8*)&smem[
-                                    row * EPI2_ROW + koff];
-                            }
-#pragma unroll
-                            for (int j = 0; j < 4; j++) {
-                                int row = wcol64 + j * 16 + lrow;
-                                int off = (ss * 32 + kq * 8)
-                                          ^ (swz_row(row) << 3);
-                                vbf[ss][j] = *(const short8*)&Vt[
-                                    row * FBK + off];
-                            }
-                        }
-#pragma unroll
-                        for (int ss = 0; ss < 2; ss++)
-#pragma unroll
-                            for (int i = 0; i < 4; i++)
-#pragma unroll
-                                for (int j = 0; j < 4; j++)
-                                    acc2[i][j] =
-                                        __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                            paf[ss][i], vbf[ss][j],
-                                            acc2[i][j], 0, 0, 0);
-                        __syncthreads();
                     }
-                    // store O block: rows m0+wrow.., cols dc+wcol64..
+                };
+                auto write_piece = [&](int st) {
+                    if (tt < 256) {
+                        ushort_t* Vt = (st & 1) ? Vt1 : Vt0;
 #pragma unroll
-                    for (int i16 = 0; i16 < 4; i16++)
+                        for (int e = 0; e < 8; e++) {
+                            Pc col;
 #pragma unroll
-                        for (int r = 0; r < 4; r++) {
-                            long oi = m0 + wrow + i16 * 16 + kq * 4 + r;
-                            ushort_t* orow2 = Op + oi * p.out2_ld;
-#pragma unroll
-                            for (int j16 = 0; j16 < 4; j16++) {
-                                int oc = dc + wcol64 + j16 * 16 + lrow;
-                                if (oc < dtot)
-                                    orow2[oc] = f2bf(acc2[i16][j16][r]);
-                            }
+                            for (int r = 0; r < 8; r++)
+                                col.u[r] = rv[r].u[e];
+                            int row = cb * 8 + e;
+                            int off = (mb * 8) ^ (swz_row(row) << 3);
+                            *(uint4v*)&Vt[row * FBK + off] = col.v;
                         }
+                    }
+                };
+                f32x4 acc2[4][4] = {};
+                load_piece(0);
+                write_piece(0);
+                __syncthreads();
+                for (int st = 0; st < nstage; st++) {
+                    if (st + 1 < nstage) load_piece(st + 1);
+                    const ushort_t* Vt = (st & 1) ? Vt1 : Vt0;
+                    const int jt = st & 3, dc = (st >> 2) * 256;
+                    short8 paf[2][4], vbf[2][4];
+#pragma unroll
+                    for (int ss = 0; ss < 2; ss++) {
+#pragma unroll
+                        for (int i = 0; i < 4; i++) {
+                            int row = wrow + i * 16 + lrow;
+                            int koff = jt * 64 + ss * 32 + kq * 8;
+                            paf[ss][i] = *(const short8*)&smem[
+                                row * EPI2_ROW + koff];
+                        }
+#pragma unroll
+                        for (int j = 0; j < 4; j++) {
+                            int row = wcol64 + j * 16 + lrow;
+                            int off = (ss * 32 + kq * 8)
+                                      ^ (swz_row(row) << 3);
+                            vbf[ss][j] = *(const short8*)&Vt[
+                                row * FBK + off];
+                        }
+                    }
+#pragma unroll
+                    for (int ss = 0; ss < 2; ss++)
+#pragma unroll
+                        for (int i = 0; i < 4; i++)
+#pragma unroll
+                            for (int j = 0; j < 4; j++)
+                                acc2[i][j] =
+                                    __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                        paf[ss][i], vbf[ss][j],
+                                        acc2[i][j], 0, 0, 0);
+                    if (st + 1 < nstage) write_piece(st + 1);
+                    if ((st & 3) == 3) {
+                        // dc chunk complete: store O from registers
+#pragma unroll
+                        for (int i16 = 0; i16 < 4; i16++)
+#pragma unroll
+                            for (int r = 0; r < 4; r++) {
+                                long oi = m0 + wrow + i16 * 16 + kq * 4 + r;
+                                ushort_t* orow2 = Op + oi * p.out2_ld;
+#pragma unroll
+                                for (int j16 = 0; j16 < 4; j16++) {
+                                    int oc = dc + wcol64 + j16 * 16 + lrow;
+                                    if (oc < dtot)
+                                        orow2[oc] = f2bf(acc2[i16][j16][r]);
+                                    acc2[i16][j16][r] = 0.f;
+                                }
+                            }
+                    }
+                    __syncthreads();
                 }
             }
         } else {
