@@ -35,6 +35,10 @@ def parse_args():
                    default="native",
                    help="native = CDNA4 HIP engine; eager/compile = stock "
                         "PyTorch-ROCm running the same math (comparison)")
+    p.add_argument("--micro", type=int, default=1,
+                   help="pipelined microbatches per step (chunk i+1's "
+                        "forward overlaps chunk i's backward; gradients "
+                        "are exactly the full-batch gradient)")
     p.add_argument("--bucket-mb", type=int, default=16,
                    help="DP gradient all-reduce bucket size (MB); tune "
                         "against the 7-link xGMI ring bandwidth")
@@ -90,7 +94,8 @@ def main():
             return 0.0
     else:
         trainer = DenoisingTrainer(model, distributed=distributed,
-                                   bucket_bytes=args.bucket_mb << 20)
+                                   bucket_bytes=args.bucket_mb << 20,
+                                   micro_batches=args.micro)
 
         def step(x, iters):
             # loss stays on-device inside the timed loop (no host sync);

@@ -68,7 +68,16 @@ class BucketedDDP:
         for bi, b in enumerate(self.buckets):
             self._pending[bi] = len(b)
 
+    def set_accumulate(self, on: bool):
+        """Gradient-accumulation window (the DDP no_sync pattern): while
+        on, backward hooks do NOT launch all-reduces — grads accumulate
+        locally. Turn off before the LAST microbatch's backward; that
+        backward then reduces the accumulated sums."""
+        self._accumulate = on
+
     def _on_grad(self, p):
+        if getattr(self, "_accumulate", False):
+            return
         bi = self._bucket_of[p]
         # Exactly one backward per finalize() is supported: a second
         # backward before finalize() would re-reduce stale flats and

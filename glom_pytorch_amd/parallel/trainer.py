@@ -39,7 +39,7 @@ class DenoisingTrainer:
     def __init__(self, model, *, lr=3e-4, noise_std=1.0, decode_step=7,
                  grad_clip=1.0, distributed=False, bucket_bytes=16 << 20,
                  process_group=None, graph_step=True, overlap_tail=True,
-                 log_path=None):
+                 micro_batches=1, log_path=None):
         """``process_group``: the group whose ranks are data-parallel
         replicas (default: the global group). For 2-D DP x SP meshes pass
         the DP subgroup here — BucketedDDP AVERAGES over its group, which
@@ -98,6 +98,14 @@ class DenoisingTrainer:
         self.overlap_tail = (overlap_tail
                              and os.environ.get("GLOM_NO_OVERLAP_TAIL",
                                                 "0") != "1")
+        # pipelined microbatching: split the batch in `micro_batches`
+        # chunks; each chunk's loss is scaled 1/m so the summed grads
+        # equal the full-batch gradient EXACTLY (mean-MSE is linear in
+        # batch averaging); chunk i+1's forward runs on a side stream
+        # overlapping chunk i's backward.
+        self.micro_batches = int(os.environ.get("GLOM_MICRO_BATCHES",
+                                                str(micro_batches)))
+        self._micro_stream = None
         self._graphs: dict = {}
 
     @staticmethod
@@ -129,6 +137,10 @@ class DenoisingTrainer:
                 and os.environ.get("GLOM_FORCE_EAGER", "0") != "1")
 
     def _eager_step(self, img: torch.Tensor, iters: int) -> torch.Tensor:
+        m = self.micro_batches
+        if (m > 1 and img.is_cuda and img.shape[0] % m == 0
+                and not getattr(self.model, "force_eager", False)):
+            return self._micro_step(img, iters, m)
         t = min(self.decode_step, iters)
         self.opt.zero_grad(set_to_none=True)
         noised = img + torch.randn_like(img) * self.noise_std
@@ -171,6 +183,76 @@ class DenoisingTrainer:
                 torch.nn.utils.clip_grad_norm_(
                     [q for g in self.opt.param_groups for q in g["params"]],
                     self.grad_clip)
+            self.opt.step()
+        return loss.detach()
+
+    def _micro_forward(self, half: torch.Tensor, iters: int, t: int,
+                       scale: float) -> torch.Tensor:
+        noised = half + torch.randn_like(half) * self.noise_std
+        traj = self.model(noised, iters=iters, return_all=True,
+                          grad_iters=t, overlap_tail=True)
+        recon = self.decoder(traj[t, :, :, -1])
+        return F.mse_loss(recon.float(), half.float()) * scale
+
+    def _micro_step(self, img: torch.Tensor, iters: int,
+                    m: int) -> torch.Tensor:
+        """Pipelined microbatches: forward of chunk i+1 (on a side
+        stream) overlaps backward of chunk i. Gradients are EXACTLY the
+        full-batch gradient: mean-MSE over the batch = (1/m) sum of the
+        chunk means, and autograd sums the scaled chunk grads."""
+        t = min(self.decode_step, iters)
+        self.opt.zero_grad(set_to_none=True)
+        if self._micro_stream is None:
+            self._micro_stream = torch.cuda.Stream()
+        s_f = self._micro_stream
+        halves = img.chunk(m)
+        if self.distributed:
+            self.ddp_model.set_accumulate(True)
+            self.ddp_dec.set_accumulate(True)
+        cur = torch.cuda.current_stream()
+        losses = []
+        pending = self._micro_forward(halves[0], iters, t, 1.0 / m)
+        for i in range(1, m):
+            # fork chunk i's forward before launching chunk i-1's backward
+            ev = torch.cuda.Event()
+            ev.record(cur)
+            with torch.cuda.stream(s_f):
+                s_f.wait_event(ev)
+                nxt = self._micro_forward(halves[i], iters, t, 1.0 / m)
+            pending.backward()
+            losses.append(pending.detach())
+            if self.distributed and i == m - 1:
+                self.ddp_model.set_accumulate(False)
+                self.ddp_dec.set_accumulate(False)
+                # counts were consumed? hooks were suppressed: reset
+                self.ddp_model._reset_counts()
+                self.ddp_dec._reset_counts()
+            pending = nxt
+        pending.backward()   # engine runs this chunk's ops on s_f
+        losses.append(pending.detach())
+        cur.wait_stream(s_f)
+        if self.distributed:
+            self.ddp_model.finalize()
+            self.ddp_dec.finalize()
+        from glom_pytorch_amd.ops.functional import join_tail_stream
+        join_tail_stream()
+        loss = losses[0]
+        for l in losses[1:]:
+            loss = loss + l
+        if self.fused_opt is not None:
+            self.fused_opt.step()
+        elif self.master is not None:
+            with torch.no_grad():
+                for mw, q in zip(self.master, self._params):
+                    mw.grad = (q.grad.float() if q.grad is not None
+                               else None)
+                if self.grad_clip:
+                    torch.nn.utils.clip_grad_norm_(self.master,
+                                                   self.grad_clip)
+                self.opt.step()
+                for mw, q in zip(self.master, self._params):
+                    q.data.copy_(mw)
+        else:
             self.opt.step()
         return loss.detach()
 

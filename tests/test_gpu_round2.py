@@ -288,3 +288,24 @@ def test_overlap_tail_matches_sequential():
     for (na, pa), (nb, pb) in zip(ta.model.named_parameters(),
                                   tb.model.named_parameters()):
         assert torch.equal(pa, pb), na
+
+
+def test_micro_pipeline_matches_full_batch():
+    """Pipelined microbatching (chunk i+1 forward overlapping chunk i
+    backward, losses scaled 1/m) must produce the full-batch gradients:
+    mean-MSE over the batch equals the mean of the chunk means."""
+    img = torch.randn(8, 3, 32, 32, device=DEV).to(torch.bfloat16)
+    tf = _trainer(graph=False)
+    tm = _trainer(graph=False, micro_batches=2)
+    tm.model.load_state_dict(tf.model.state_dict())
+    tm.decoder.load_state_dict(tf.decoder.state_dict())
+    with torch.no_grad():
+        for mw, q in zip(tm.master, tm._params):
+            mw.copy_(q.float())
+    lf = [tf.step(img, iters=4) for _ in range(3)]
+    lm = [tm.step(img, iters=4) for _ in range(3)]
+    for a, b in zip(lf, lm):
+        assert abs(a - b) < 2e-2 * max(1.0, abs(a)), (a, b)
+    for (nf, pf), (nm, pm) in zip(tf.model.named_parameters(),
+                                  tm.model.named_parameters()):
+        assert _cos(pf, pm) > 0.999, (nf, _cos(pf, pm))
