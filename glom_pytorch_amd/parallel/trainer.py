@@ -131,7 +131,11 @@ class DenoisingTrainer:
         return loss.item() if sync_loss else loss.detach()
 
     def _graph_usable(self, img) -> bool:
-        return (self.graph_step and self.fused_opt is not None
+        # micro-pipelining accumulates grads across streams; the
+        # AccumulateGrad node's stream mismatch breaks hipGraph capture
+        # (torch warns, replay faults) — micro mode runs eager-launched
+        return (self.graph_step and self.micro_batches == 1
+                and self.fused_opt is not None
                 and img.is_cuda and img.dtype == torch.bfloat16
                 and not getattr(self.model, "force_eager", False)
                 and os.environ.get("GLOM_FORCE_EAGER", "0") != "1")
