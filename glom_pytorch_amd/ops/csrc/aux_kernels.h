@@ -17,10 +17,8 @@ void launch_knorm_combine(const void* dkhat, const void* lev,
 void launch_mix_fwd(const void* prev, const void* bu, const void* td,
                     const void* cons, void* out, long total, int L, int d,
                     hipStream_t s);
-void launch_mix_bwd(const void* dout, const void* dout2, void* dmix,
-                    void* dtd, long total, int L, int d, hipStream_t s);
+void launch_mix_bwd(const void* dout, void* dmix, void* dtd, long total,
+                    int L, int d, hipStream_t s);
 void launch_add4(const void* a, const void* b, const void* c, const void* d,
                  void* out, long total, hipStream_t s);
-void launch_addn(const void* const* ins, int n, void* out, long total,
-                 int acc, hipStream_t s);
 void launch_gelu(const void* in, void* out, long total, hipStream_t s);

@@ -291,29 +291,21 @@ __global__ __launch_bounds__(NTHREADS) void k_mix_fwd(
     *(uint4v*)(out + idx) = o.v;
 }
 
-// dmix = dout / c_l (shared by prev/bu/cons); dtd = dout[..., :L-1, :] / 4.
-// dout2 (optional): second incoming gradient summed on load — fuses the
-// loop-level carry add (trajectory grad + next-step carry) for free.
+// dmix = dout / c_l (shared by prev/bu/cons); dtd = dout[..., :L-1, :] / 4
 __global__ __launch_bounds__(NTHREADS) void k_mix_bwd(
-        const ushort_t* __restrict__ dout, const ushort_t* __restrict__ dout2,
-        ushort_t* __restrict__ dmix,
+        const ushort_t* __restrict__ dout, ushort_t* __restrict__ dmix,
         ushort_t* __restrict__ dtd, long total, int L, int d) {
     long idx = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
     if (idx >= total) return;
     int q = idx % d;
     int l = (idx / d) % L;
     long bn = idx / ((long)d * L);
-    union { uint4v v; ushort_t u[8]; } g, g2, o;
+    union { uint4v v; ushort_t u[8]; } g, o;
     g.v = *(const uint4v*)(dout + idx);
-    if (dout2) g2.v = *(const uint4v*)(dout2 + idx);
     bool has_td = l < L - 1;
     float c = has_td ? 4.0f : 3.0f;
 #pragma unroll
-    for (int e = 0; e < 8; e++) {
-        float v = bf2f(g.u[e]);
-        if (dout2) v += bf2f(g2.u[e]);
-        o.u[e] = f2bf(v / c);
-    }
+    for (int e = 0; e < 8; e++) o.u[e] = f2bf(bf2f(g.u[e]) / c);
     *(uint4v*)(dmix + idx) = o.v;
     if (has_td) *(uint4v*)(dtd + (bn * (L - 1) + l) * (long)d + q) = o.v;
 }
@@ -397,11 +389,11 @@ void launch_mix_fwd(const void* prev, const void* bu, const void* td,
                        (ushort_t*)out, total, L, d);
 }
 
-void launch_mix_bwd(const void* dout, const void* dout2, void* dmix,
-                    void* dtd, long total, int L, int d, hipStream_t s) {
+void launch_mix_bwd(const void* dout, void* dmix, void* dtd, long total,
+                    int L, int d, hipStream_t s) {
     hipLaunchKernelGGL(k_mix_bwd, dim3(cdiv(total / 8, NTHREADS)), dim3(NTHREADS),
-                       0, s, (const ushort_t*)dout, (const ushort_t*)dout2,
-                       (ushort_t*)dmix, (ushort_t*)dtd, total, L, d);
+                       0, s, (const ushort_t*)dout, (ushort_t*)dmix,
+                       (ushort_t*)dtd, total, L, d);
 }
 
 // out = a + b + c + d, elementwise bf16 (the four levels-gradient
@@ -455,41 +447,4 @@ void launch_gelu(const void* in, void* out, long total, hipStream_t s) {
     hipLaunchKernelGGL(k_gelu, dim3(cdiv(total / 8, NTHREADS)),
                        dim3(NTHREADS), 0, s, (const ushort_t*)in,
                        (ushort_t*)out, total);
-}
-
-// out (+)= a0 + a1 + ... + a_{n-1}, up to 8 inputs per launch; acc=1 adds
-// into the existing out (chaining for >8 summands). Replaces the autograd
-// engine's pairwise gradient adds with one pass.
-struct AddNArgs { const ushort_t* in[8]; };
-__global__ __launch_bounds__(NTHREADS) void k_addn(
-        AddNArgs a, int n, ushort_t* __restrict__ out, long total, int acc) {
-    long i8 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 8;
-    if (i8 >= total) return;
-    float v[8] = {};
-    if (acc) {
-        union { uint4v w; ushort_t u[8]; } t;
-        t.w = *(const uint4v*)(out + i8);
-#pragma unroll
-        for (int e = 0; e < 8; e++) v[e] = bf2f(t.u[e]);
-    }
-    for (int j = 0; j < n; j++) {
-        union { uint4v w; ushort_t u[8]; } t;
-        t.w = *(const uint4v*)(a.in[j] + i8);
-#pragma unroll
-        for (int e = 0; e < 8; e++) v[e] += bf2f(t.u[e]);
-    }
-    union { uint4v w; ushort_t u[8]; } o;
-#pragma unroll
-    for (int e = 0; e < 8; e++) o.u[e] = f2bf(v[e]);
-    *(uint4v*)(out + i8) = o.w;
-}
-
-void launch_addn(const void* const* ins, int n, void* out, long total,
-                 int acc, hipStream_t s) {
-    AddNArgs a;
-    for (int j = 0; j < n; j++) a.in[j] = (const ushort_t*)ins[j];
-    for (int j = n; j < 8; j++) a.in[j] = nullptr;
-    hipLaunchKernelGGL(k_addn, dim3(cdiv(total / 8, NTHREADS)),
-                       dim3(NTHREADS), 0, s, a, n, (ushort_t*)out, total,
-                       acc);
 }

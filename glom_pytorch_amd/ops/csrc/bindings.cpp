@@ -650,51 +650,19 @@ torch::Tensor level_mix_fwd(torch::Tensor levels, torch::Tensor bu,
     return out;
 }
 
-std::vector<torch::Tensor> level_mix_bwd(
-        torch::Tensor dout,
-        c10::optional<torch::Tensor> dout2 = c10::nullopt) {
+std::vector<torch::Tensor> level_mix_bwd(torch::Tensor dout) {
     CHECK_IN(dout);
     const int64_t B = dout.size(0), N = dout.size(1), L = dout.size(2),
                   d = dout.size(3);
     TORCH_CHECK(d % 8 == 0, "dim must be a multiple of 8");
     auto dmix = torch::empty_like(dout);
     auto dtd = torch::empty({B, N, L - 1, d}, dout.options());
-    const void* d2 = nullptr;
-    if (dout2.has_value() && dout2->numel() > 0) {
-        CHECK_IN((*dout2));
-        d2 = dout2->data_ptr();
-    }
-    launch_mix_bwd(dout.data_ptr(), d2, dmix.data_ptr(), dtd.data_ptr(),
+    launch_mix_bwd(dout.data_ptr(), dmix.data_ptr(), dtd.data_ptr(),
                    dout.numel(), (int)L, (int)d, cur_stream());
     check_launch();
     return {dmix, dtd};
 }
 
-// out = sum(inputs) (all bf16, same numel%8==0); one fused pass replacing
-// the autograd engine's pairwise grad accumulation adds.
-torch::Tensor addn(std::vector<torch::Tensor> ins) {
-    TORCH_CHECK(!ins.empty());
-    auto out = torch::empty_like(ins[0]);
-    long total = out.numel();
-    TORCH_CHECK(total % 8 == 0, "addn needs numel %% 8 == 0");
-    hipStream_t s = cur_stream();
-    const void* ptrs[8];
-    size_t i = 0;
-    int acc = 0;
-    while (i < ins.size()) {
-        int n = (int)std::min<size_t>(8, ins.size() - i);
-        for (int j = 0; j < n; j++) {
-            CHECK_IN(ins[i + j]);
-            TORCH_CHECK(ins[i + j].numel() == total);
-            ptrs[j] = ins[i + j].data_ptr();
-        }
-        launch_addn(ptrs, n, out.data_ptr(), total, acc, s);
-        check_launch();
-        acc = 1;
-        i += n;
-    }
-    return out;
-}
 
 
 // ------------------------------------------------------------------ //
@@ -1147,9 +1115,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("level_mix_fwd", &level_mix_fwd, "level mix forward",
           py::arg("levels"), py::arg("bu"), py::arg("td"), py::arg("cons"),
           py::arg("slab") = c10::nullopt, py::arg("slab_idx") = 0);
-    m.def("level_mix_bwd", &level_mix_bwd, "level mix backward",
-          py::arg("dout"), py::arg("dout2") = c10::nullopt);
-    m.def("addn", &addn, "fused N-way bf16 sum");
+    m.def("level_mix_bwd", &level_mix_bwd, "level mix backward");
     m.def("glom_step_fwd", &glom_step_fwd, "full GLOM iteration forward",
           py::arg("tokens"), py::arg("levels"), py::arg("pos"),
           py::arg("bw1"), py::arg("bb1"), py::arg("bw2"), py::arg("bb2"),
