@@ -18,6 +18,11 @@ where it received a gradient; this class uses the global step count for
 every present param's bias correction (params with ``grad=None`` are
 skipped entirely, like torch). Identical whenever all params get grads
 every step — the standard training path.
+
+Hyperparameter caveat under graph capture: ``lr``/betas/eps/weight_decay
+are kernel arguments, so a captured training step bakes them in. To
+change them mid-run, update the attribute AND drop the captured graphs
+(``trainer._graphs.clear()``) so the next step re-captures.
 """
 
 from __future__ import annotations
