@@ -79,3 +79,50 @@ def test_embed_rejects_malformed_body():
         r = client.post("/embed", content=b"\x00" * 100)
         assert r.status_code == 400
         assert b"expected" in r.content
+
+
+def test_healthz_live_during_slow_inference():
+    """The forward runs in a worker thread (run_in_executor), so /healthz
+    answers while a request is being served — the liveness probe measures
+    liveness, not inference latency (round-1 ADVICE item)."""
+    import threading
+    import time
+
+    torch.manual_seed(0)
+    model = Glom(**SMALL)
+    size = model.image_size
+
+    slow_gate = threading.Event()
+
+    class SlowGlom(torch.nn.Module):
+        def __init__(self, inner):
+            super().__init__()
+            self.inner = inner
+            self.image_size = inner.image_size
+
+        def parameters(self, *a, **k):
+            return self.inner.parameters(*a, **k)
+
+        def forward(self, *a, **k):
+            slow_gate.wait(timeout=5.0)   # block the WORKER thread only
+            return self.inner(*a, **k)
+
+    app = create_app(SlowGlom(model), iters=2, max_wait_ms=0.5)
+    with TestClient(app) as client:
+        body = np.zeros((3, size, size), dtype=np.float32).tobytes()
+        results = {}
+
+        def post():
+            results["embed"] = client.post("/embed", content=body)
+
+        th = threading.Thread(target=post)
+        th.start()
+        time.sleep(0.3)    # request is now blocked inside the worker
+        t0 = time.perf_counter()
+        h = client.get("/healthz")
+        dt = time.perf_counter() - t0
+        assert h.status_code == 200 and h.json()["ok"]
+        assert dt < 1.0, f"healthz took {dt:.2f}s while inference ran"
+        slow_gate.set()
+        th.join(timeout=10)
+        assert results["embed"].status_code == 200
