@@ -30,7 +30,7 @@ ext = CUDAExtension(
 
 setup(
     name="glom_pytorch_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=["glom_pytorch_amd", "glom_pytorch_amd.models",
               "glom_pytorch_amd.ops", "glom_pytorch_amd.parallel",
               "glom_pytorch_amd.utils"],
