@@ -108,3 +108,27 @@ def test_nt5p_dispatches_on_headline_shapes():
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-800:]
     assert "nt5p=1" in r.stderr, r.stderr[-800:]
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_nt8p_dispatches_on_down_projection():
+    """The 8-phase 256^2 kernel must own the K>=2048 NT shapes (default
+    on since the overlap schedule flipped its e2e sign) — and must NOT
+    take the K=512 up-projection family."""
+    import os
+    import subprocess
+    import sys
+    env = dict(os.environ, GLOM_DISPATCH_DEBUG="1")
+    code = ("import torch; from glom_pytorch_amd.ops import _load_extension;"
+            " ext=_load_extension();"
+            " ext.bench_gemm(16384,512,2048,0,6,0,1);"     # down-proj
+            " ext.bench_gemm(16384,2048,512,0,6,0,1)")     # up-proj
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    lines = [l for l in r.stderr.splitlines() if "[dispatch]" in l]
+    down = [l for l in lines if "N512 K2048" in l]
+    up = [l for l in lines if "N2048 K512" in l]
+    assert down and "nt8p=1" in down[0], down
+    assert up and "nt8p=0" in up[0] and "nt5p=1" in up[0], up
