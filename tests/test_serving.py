@@ -126,3 +126,27 @@ def test_healthz_live_during_slow_inference():
         slow_gate.set()
         th.join(timeout=10)
         assert results["embed"].status_code == 200
+
+
+def test_microbatcher_pads_to_power_of_two():
+    """Batch shapes are padded to the next power of two so hipGraph
+    capture is reused across a handful of shapes instead of one graph
+    per batch size."""
+    torch.manual_seed(0)
+    model = Glom(**SMALL)
+    from glom_pytorch_amd.serving import MicroBatcher
+    mb = MicroBatcher(model, iters=2)
+    seen = []
+    orig_forward = model.forward
+
+    def spy(img, *a, **k):
+        seen.append(img.shape[0])
+        return orig_forward(img, *a, **k)
+
+    model.forward = spy
+    size = model.image_size
+    imgs = [torch.randn(3, size, size) for _ in range(3)]
+    outs = mb._run(imgs)
+    assert seen == [4]            # 3 requests -> padded to 4
+    assert len(outs) == 3         # padding rows dropped from results
+    assert outs[0].shape == (model.num_patches, model.dim)
