@@ -21,6 +21,11 @@ model = Glom(dim=512 if use_gpu else 64, levels=6 if use_gpu else 3,
              image_size=size, patch_size=patch).to(dev, dtype)
 
 img = torch.randn(2, 3, size, size, device=dev, dtype=dtype)
+# Tip: when your loss only reads trajectory times <= t, pass
+# grad_iters=t — later iterations then run forward-only (their gradient
+# contribution is exactly zero) and, inside DenoisingTrainer, overlap
+# the backward. This loss reads the FINAL state, so the full graph is
+# kept here.
 all_levels = model(img, iters=2 * model.levels, return_all=True)
 print("trajectory:", tuple(all_levels.shape))
 
