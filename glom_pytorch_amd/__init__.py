@@ -16,6 +16,6 @@ A from-scratch reimplementation of the capabilities of lucidrains/glom-pytorch
 
 from glom_pytorch_amd.models.glom import Glom, GroupedFeedForward, ConsensusAttention
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = ["Glom", "GroupedFeedForward", "ConsensusAttention"]
